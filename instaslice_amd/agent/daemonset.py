@@ -1,0 +1,386 @@
+"""L2 node agent: realizes partitions on the GPUs via the amd-smi layer.
+
+Reference analog: InstaSliceDaemonsetReconciler
+(instaslice_daemonset.go:95-275) with the call stacks of SURVEY.md §3.2/§3.3.
+Responsibilities:
+
+  boot      discover GPUs + profile catalog ONCE (cached — fixes the
+            reference's per-reconcile nvml.Init hot spot, :112), publish the
+            node's Instaslice CR, adopt pre-existing ("dangling") partitions
+  creating  ensure the GPU is in the allocation's compute/memory mode
+            (whole-GPU set; idle-checked), resolve the partition device at the
+            allocated ordinal, publish the pod's visible-devices ConfigMap,
+            patch node capacity, record Prepared, flip status -> created
+  deleted   drop ConfigMap + capacity + Prepared, release the ordinal,
+            remove the allocation entry
+
+Every reconfigure captures amd-smi metrics before/after plus the mode-set
+wall time (the north-star observability requirement; the reference registers
+no custom metrics).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional
+
+from instaslice_amd import POD_RESOURCE_PREFIX
+from instaslice_amd.api.types import (
+    AllocationStatus,
+    GpuStatus,
+    PreparedDetails,
+    new_instaslice,
+)
+from instaslice_amd.partition.profiles import catalog_from_amdsmi_profiles
+from instaslice_amd.runtime.engine import Engine, Key, Result, WatchSpec
+from instaslice_amd.smi.base import AmdSmi, PhysicalGpu, SmiBusy, SmiError, SmiNotSupported
+from instaslice_amd.store.memstore import AlreadyExists, MemStore, NotFound
+from instaslice_amd.utils import get_logger
+
+INSTASLICE_NS = "instaslice-system"
+REQUEUE_BUSY_S = 0.5
+
+
+class NodeAgent:
+    def __init__(
+        self,
+        store: MemStore,
+        smi: AmdSmi,
+        node_name: str,
+        reset_mode_on_empty: bool = False,
+    ) -> None:
+        self.store = store
+        self.smi = smi
+        self.node_name = node_name
+        # The reference destroys each slice on teardown; AMD mode flips are
+        # whole-GPU and comparatively expensive, so by default the mode is
+        # sticky when a GPU drains (the next same-profile pod lands with zero
+        # reconfiguration). Set True for reference-parity teardown.
+        self.reset_mode_on_empty = reset_mode_on_empty
+        self.log = get_logger(f"agent.{node_name}")
+        # cached enumeration: gpu uuid -> PhysicalGpu (refreshed only after a
+        # mode change of that GPU)
+        self._gpus: Dict[str, PhysicalGpu] = {}
+        # observability feed (metrics module scrapes this)
+        self.reconfigure_events: List[dict] = []
+        self.engine = Engine(
+            name=f"agent-{node_name}",
+            store=store,
+            reconcile=self._reconcile,
+            watches=[WatchSpec(kind="Instaslice", map_fn=self._own_node_only)],
+        )
+
+    def _own_node_only(self, event_type: str, obj: dict) -> List[Key]:
+        if obj["metadata"]["name"] != self.node_name:
+            return []
+        return [("Instaslice", obj["metadata"].get("namespace", ""), self.node_name)]
+
+    # -- discovery (boot, once) --------------------------------------------
+    # Reference: discoverMigEnabledGpuWithSlices + discoverAvailableProfilesOnGpus
+    # + discoverDanglingSlices (instaslice_daemonset.go:555-748).
+
+    def discover(self) -> dict:
+        self.smi.init()
+        gpus = self.smi.list_gpus()
+        self._gpus = {g.uuid: g for g in gpus}
+        if not gpus:
+            raise SmiError(f"node {self.node_name}: no AMD GPUs found")
+        first = gpus[0]
+        try:
+            raw_profiles = self.smi.get_profile_config(first.uuid)
+        except (SmiNotSupported, SmiError) as e:
+            self.log.warning("profile discovery unsupported (%s); using MI355X model", e)
+            raw_profiles = []
+        catalog = catalog_from_amdsmi_profiles(
+            first.model, first.memory_gb, raw_profiles
+        )
+
+        cr = new_instaslice(self.node_name)
+        spec = cr["spec"]
+        for g in gpus:
+            spec["gpuUuids"][g.uuid] = g.model
+            spec["gpus"][g.uuid] = {
+                **GpuStatus(
+                    uuid=g.uuid,
+                    model=g.model,
+                    memory_gb=g.memory_gb,
+                    compute_mode=g.compute_mode,
+                    memory_mode=g.memory_mode,
+                ).to_dict(),
+                "index": g.index,
+            }
+        spec["placements"] = catalog.to_dict()
+        cr["status"]["processed"] = "true"
+
+        try:
+            existing = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+        except NotFound:
+            existing = None
+
+        if existing is None:
+            try:
+                self.store.create(cr)
+            except AlreadyExists:
+                existing = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+        if existing is not None:
+            # agent restart: adopt durable state (allocations/prepared survive
+            # in the CR — reference skips re-discovery when Processed=="true",
+            # instaslice_daemonset.go:528-534 — but we still refresh live modes)
+            def refresh(obj: dict) -> Optional[dict]:
+                obj["spec"]["gpuUuids"] = spec["gpuUuids"]
+                obj["spec"]["placements"] = spec["placements"]
+                for uuid, gd in spec["gpus"].items():
+                    old = obj["spec"].setdefault("gpus", {}).get(uuid)
+                    if old:
+                        gd = dict(gd, usedOrdinals=old.get("usedOrdinals", []))
+                    obj["spec"]["gpus"][uuid] = gd
+                obj["status"]["processed"] = "true"
+                return obj
+
+            self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, refresh)
+
+        # publish the Node object (capacity patches land on it)
+        node = {
+            "apiVersion": "v1",
+            "kind": "Node",
+            "metadata": {"name": self.node_name, "namespace": "", "labels": {}},
+            "status": {"capacity": {}},
+        }
+        try:
+            self.store.create(node)
+        except AlreadyExists:
+            pass
+        return self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+
+    # -- mode management ----------------------------------------------------
+
+    def _ensure_gpu_mode(self, cr: dict, gpu_uuid: str, compute: str, memory: str) -> None:
+        """Whole-GPU mode set with metrics capture + cached re-enumeration.
+        Raises SmiBusy if the device refuses (caller requeues)."""
+        g = self._gpus.get(gpu_uuid)
+        if g is None:
+            raise SmiError(f"unknown gpu {gpu_uuid}")
+        changed = False
+        event = {
+            "gpu": gpu_uuid,
+            "node": self.node_name,
+            "from": f"{g.compute_mode}/{g.memory_mode}",
+            "to": f"{compute}/{memory}",
+        }
+        try:
+            event["metrics_before"] = self.smi.get_metrics(gpu_uuid)
+        except SmiError:
+            event["metrics_before"] = {}
+        t0 = time.monotonic()
+        if g.compute_mode != compute:
+            self.smi.set_compute_partition(gpu_uuid, compute)
+            changed = True
+        if g.memory_mode != memory:
+            try:
+                self.smi.set_memory_partition(gpu_uuid, memory)
+                changed = True
+            except SmiNotSupported:
+                # NPS flip may be impossible (VM guest / needs driver reload):
+                # capacity math is unaffected, only locality — log and go on.
+                self.log.warning(
+                    "gpu %s: memory mode %s unsupported, staying on %s",
+                    gpu_uuid[:8], memory, g.memory_mode,
+                )
+        event["set_wall_ms"] = (time.monotonic() - t0) * 1000.0
+        if changed:
+            # CPX re-enumeration: one handle becomes N
+            for fresh in self.smi.list_gpus():
+                self._gpus[fresh.uuid] = fresh
+            try:
+                event["metrics_after"] = self.smi.get_metrics(gpu_uuid)
+            except SmiError:
+                event["metrics_after"] = {}
+            self.reconfigure_events.append(event)
+            g = self._gpus[gpu_uuid]
+            self.log.info(
+                "gpu %s reconfigured %s -> %s/%s in %.1f ms (%d partitions)",
+                gpu_uuid[:8], event["from"], g.compute_mode, g.memory_mode,
+                event["set_wall_ms"], len(g.partitions),
+            )
+
+    # -- create path ----------------------------------------------------------
+
+    def _prepare_allocation(self, cr: dict, alloc: dict) -> Optional[dict]:
+        """Realize one `creating` allocation; returns the prepared-entry dict
+        keyed by partition uuid, or None if it must be retried later."""
+        gpu_uuid = alloc["gpuUUID"]
+        self._ensure_gpu_mode(cr, gpu_uuid, alloc["computeMode"], alloc["memoryMode"])
+        g = self._gpus[gpu_uuid]
+        ordinal = alloc["ordinal"]
+        part = next((p for p in g.partitions if p.ordinal == ordinal), None)
+        if part is None:
+            raise SmiError(
+                f"gpu {gpu_uuid}: ordinal {ordinal} not present in mode {g.compute_mode}"
+            )
+        # the workload contract: a pod-named ConfigMap delivering the visible
+        # partition (reference: createConfigMap with NVIDIA_/CUDA_VISIBLE_DEVICES,
+        # instaslice_daemonset.go:796-818 -> ROCR_/HIP_ here)
+        cm = {
+            "apiVersion": "v1",
+            "kind": "ConfigMap",
+            "metadata": {"name": alloc["podName"], "namespace": alloc["namespace"]},
+            "data": {
+                "ROCR_VISIBLE_DEVICES": part.uuid,
+                "HIP_VISIBLE_DEVICES": str(part.device_index),
+                "INSTASLICE_PARTITION_ORDINAL": str(ordinal),
+                "INSTASLICE_PARTITION_GPU": gpu_uuid,
+            },
+        }
+        try:
+            self.store.create(cm)
+        except AlreadyExists:
+            pass
+        # node capacity pinning (reference: createInstaSliceResource,
+        # instaslice_daemonset.go:277-300)
+        self._patch_capacity(alloc["podName"], add=True)
+        prep = PreparedDetails(
+            parent_gpu_uuid=gpu_uuid,
+            ordinal=ordinal,
+            compute_mode=g.compute_mode,
+            memory_mode=g.memory_mode,
+            xcds=alloc["size"],
+            memory_gb=part.memory_gb,
+            pod_uuid=alloc["podUUID"],
+            device_index=part.device_index,
+        )
+        return {part.uuid: prep.to_dict()}
+
+    def _patch_capacity(self, pod_name: str, add: bool) -> None:
+        def mut(node: dict) -> Optional[dict]:
+            cap = node.setdefault("status", {}).setdefault("capacity", {})
+            key = POD_RESOURCE_PREFIX + pod_name
+            if add:
+                if cap.get(key) == 1:
+                    return None
+                cap[key] = 1
+            else:
+                if key not in cap:
+                    return None
+                del cap[key]
+            return node
+
+        self.store.update_with_retry("Node", self.node_name, "", mut)
+
+    # -- delete path ----------------------------------------------------------
+
+    def _teardown_allocation(self, cr: dict, pod_uuid: str, alloc: dict) -> None:
+        try:
+            self.store.delete("ConfigMap", alloc["podName"], alloc["namespace"])
+        except NotFound:
+            pass
+        self._patch_capacity(alloc["podName"], add=False)
+
+    # -- reconcile ------------------------------------------------------------
+
+    def _reconcile(self, key: Key) -> Result:
+        try:
+            cr = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+        except NotFound:
+            return Result()
+        allocations = cr.get("spec", {}).get("allocations") or {}
+        requeue: Optional[float] = None
+
+        for pod_uuid, alloc in sorted(allocations.items()):
+            status = alloc["allocationStatus"]
+            if status == AllocationStatus.CREATING:
+                try:
+                    prepared_entry = self._prepare_allocation(cr, alloc)
+                except SmiBusy as e:
+                    self.log.warning("gpu busy, requeueing: %s", e)
+                    requeue = REQUEUE_BUSY_S
+                    continue
+                if prepared_entry is None:
+                    requeue = REQUEUE_BUSY_S
+                    continue
+                part_uuid = next(iter(prepared_entry))
+
+                def commit(obj: dict, pu=pod_uuid, pe=prepared_entry, a=alloc) -> Optional[dict]:
+                    spec = obj["spec"]
+                    cur = (spec.get("allocations") or {}).get(pu)
+                    if not cur or cur["allocationStatus"] != AllocationStatus.CREATING:
+                        return None
+                    cur["allocationStatus"] = AllocationStatus.CREATED
+                    spec.setdefault("prepared", {}).update(pe)
+                    gd = spec["gpus"][a["gpuUUID"]]
+                    g = self._gpus[a["gpuUUID"]]
+                    used = set(gd.get("usedOrdinals", []))
+                    used.add(a["ordinal"])
+                    gd.update(
+                        computeMode=g.compute_mode,
+                        memoryMode=g.memory_mode,
+                        usedOrdinals=sorted(used),
+                    )
+                    return obj
+
+                self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, commit)
+                self.log.debug(
+                    "prepared partition %s for pod %s", part_uuid[:8], alloc["podName"]
+                )
+            elif status == AllocationStatus.DELETED:
+                self._teardown_allocation(cr, pod_uuid, alloc)
+
+                def cleanup(obj: dict, pu=pod_uuid, a=alloc) -> Optional[dict]:
+                    spec = obj["spec"]
+                    if pu not in (spec.get("allocations") or {}):
+                        return None
+                    del spec["allocations"][pu]
+                    prepared = spec.get("prepared") or {}
+                    for puid in [k for k, v in prepared.items() if v["podUUID"] == pu]:
+                        del prepared[puid]
+                    gd = spec["gpus"].get(a["gpuUUID"])
+                    if gd:
+                        used = set(gd.get("usedOrdinals", []))
+                        used.discard(a["ordinal"])
+                        gd["usedOrdinals"] = sorted(used)
+                    return obj
+
+                self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, cleanup)
+                if self.reset_mode_on_empty:
+                    self._maybe_reset_gpu(alloc["gpuUUID"])
+        return Result(requeue_after=requeue)
+
+    def _maybe_reset_gpu(self, gpu_uuid: str) -> None:
+        """Reference-parity teardown (ci/gi Destroy analog,
+        instaslice_daemonset.go:377-413): return a drained GPU to SPX/NPS1."""
+        try:
+            cr = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+        except NotFound:
+            return
+        spec = cr["spec"]
+        in_use = any(
+            a["gpuUUID"] == gpu_uuid for a in (spec.get("allocations") or {}).values()
+        ) or any(
+            p["parentGpuUUID"] == gpu_uuid for p in (spec.get("prepared") or {}).values()
+        )
+        if in_use:
+            return
+        try:
+            self._ensure_gpu_mode(cr, gpu_uuid, "SPX", "NPS1")
+        except (SmiBusy, SmiNotSupported) as e:
+            self.log.warning("reset of %s skipped: %s", gpu_uuid[:8], e)
+            return
+        g = self._gpus[gpu_uuid]
+
+        def mut(obj: dict) -> Optional[dict]:
+            gd = obj["spec"]["gpus"].get(gpu_uuid)
+            if not gd:
+                return None
+            gd.update(computeMode=g.compute_mode, memoryMode=g.memory_mode)
+            return obj
+
+        self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, mut)
+
+    # -- lifecycle ------------------------------------------------------------
+
+    def start(self) -> "NodeAgent":
+        self.discover()
+        self.engine.start()
+        return self
+
+    def stop(self) -> None:
+        self.engine.stop()

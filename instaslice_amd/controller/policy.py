@@ -1,0 +1,223 @@
+"""Placement policies: where does a partition request land?
+
+Reference analog: AllocationPolicy interface + FirstFitPolicy
+(instaslice_controller.go:48-50, :436-453) and the slot bin-packer
+getStartIndexFromPreparedState (:303-384). The reference's LeftToRight/
+RightToLeft policies are empty stubs (:456-469); here both shipped policies
+are real.
+
+The MI355X semantic drives the structure: AMD partitioning
+is a whole-GPU *mode*, so a placement is either
+
+  (a) a free partition ordinal on a GPU already in the profile's compute
+      mode, or
+  (b) ordinal 0 on an *idle* GPU that will be reconfigured (compute mode set,
+      and memory mode set if valid and different from preferred) — mode
+      changes require the device idle (amdsmi.h:5781).
+
+Occupancy: a partition ordinal is occupied if any allocation (any status —
+`deleted` still occupies until the daemonset cleans it) or any prepared entry
+references it; this mirrors the reference's stale-prepared guard
+(instaslice_controller.go:198-203 / SURVEY.md §5c).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Set, Tuple
+
+from instaslice_amd.api.types import GpuStatus
+from instaslice_amd.partition.profiles import (
+    ComputeMode,
+    MemoryMode,
+    PartitionProfile,
+    VALID_MEMORY_MODES,
+)
+
+
+@dataclass(frozen=True)
+class Placement:
+    """A placement decision for one request."""
+
+    node: str
+    gpu_uuid: str
+    ordinal: int
+    compute_mode: str      # mode the GPU must be in (may require a transition)
+    memory_mode: str       # memory mode the GPU will have
+    needs_mode_change: bool
+
+
+@dataclass
+class GpuView:
+    """Controller-side view of one GPU assembled from the Instaslice CR."""
+
+    node: str
+    uuid: str
+    index: int
+    memory_gb: int
+    compute_mode: ComputeMode
+    memory_mode: MemoryMode
+    occupied: Set[int]  # partition ordinals referenced by allocations/prepared
+
+
+def build_gpu_views(node_name: str, spec: dict) -> List[GpuView]:
+    """Assemble GpuViews from an Instaslice CR spec (api.types.new_instaslice
+    shape). Sorted by physical index for deterministic first-fit.
+
+    A GPU with allocations whose target mode differs from the CR's live mode
+    is *mid-transition* (the daemonset has not flipped it yet): its effective
+    mode for placement is the allocations' target mode, so back-to-back
+    requests for the same profile pack onto it instead of each grabbing a
+    fresh idle GPU. The reference has no analog (MIG slices carve next to
+    running ones); whole-GPU modes make pending transitions first-class."""
+    occupied: Dict[str, Set[int]] = {}
+    target_modes: Dict[str, Tuple[str, str]] = {}
+    for alloc in (spec.get("allocations") or {}).values():
+        occupied.setdefault(alloc["gpuUUID"], set()).add(alloc["ordinal"])
+        if alloc.get("computeMode"):
+            target_modes[alloc["gpuUUID"]] = (
+                alloc["computeMode"],
+                alloc.get("memoryMode") or "NPS1",
+            )
+    for prep in (spec.get("prepared") or {}).values():
+        occupied.setdefault(prep["parentGpuUUID"], set()).add(prep["ordinal"])
+
+    views: List[GpuView] = []
+    for uuid, gd in (spec.get("gpus") or {}).items():
+        g = GpuStatus.from_dict(gd)
+        compute, memory = g.compute_mode, g.memory_mode
+        if uuid in target_modes:
+            compute, memory = target_modes[uuid]
+        views.append(
+            GpuView(
+                node=node_name,
+                uuid=uuid,
+                index=spec_index(spec, uuid),
+                memory_gb=g.memory_gb,
+                compute_mode=ComputeMode(compute),
+                memory_mode=MemoryMode(memory),
+                occupied=occupied.get(uuid, set()),
+            )
+        )
+    views.sort(key=lambda v: v.index)
+    return views
+
+
+def spec_index(spec: dict, uuid: str) -> int:
+    gd = (spec.get("gpus") or {}).get(uuid)
+    if gd and "index" in gd:
+        return gd["index"]
+    return 0
+
+
+def _target_memory_mode(profile: PartitionProfile, current: MemoryMode) -> Tuple[MemoryMode, bool]:
+    """Memory mode is sticky (a set requires driver reload on bare metal,
+    amdsmi.h:5861): keep the current mode if it's legal under the target
+    compute mode; only plan a change to the preferred mode when the GPU is
+    being reconfigured anyway and the current mode would be illegal."""
+    legal = VALID_MEMORY_MODES[profile.compute]
+    if current in legal:
+        return current, False
+    return (
+        profile.preferred_memory if profile.preferred_memory in legal else legal[0]
+    ), True
+
+
+class AllocationPolicy:
+    """Strategy interface (reference: instaslice_controller.go:48-50)."""
+
+    name = "base"
+
+    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+        raise NotImplementedError
+
+
+def _free_ordinal(view: GpuView, n_partitions: int) -> Optional[int]:
+    for k in range(n_partitions):
+        if k not in view.occupied:
+            return k
+    return None
+
+
+def _place_on(view: GpuView, profile: PartitionProfile, needs_change: bool) -> Optional[Placement]:
+    n = profile.partitions_per_gpu
+    if needs_change:
+        if view.occupied:
+            return None  # mode change needs an idle GPU
+        mem, _ = _target_memory_mode(profile, view.memory_mode)
+        return Placement(view.node, view.uuid, 0, profile.compute.value, mem.value, True)
+    k = _free_ordinal(view, n)
+    if k is None:
+        return None
+    return Placement(
+        view.node, view.uuid, k, profile.compute.value, view.memory_mode.value, False
+    )
+
+
+class FirstFitPolicy(AllocationPolicy):
+    """Reference-parity first-fit (FirstFitPolicy, instaslice_controller.go:436-453):
+    scan GPUs in index order; take the first free ordinal on a GPU already in
+    the right mode, else the first idle GPU (planning a mode change)."""
+
+    name = "first-fit"
+
+    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+        for v in views:
+            if v.compute_mode is profile.compute:
+                p = _place_on(v, profile, needs_change=False)
+                if p:
+                    return p
+        for v in views:
+            if v.compute_mode is not profile.compute:
+                p = _place_on(v, profile, needs_change=True)
+                if p:
+                    return p
+        return None
+
+
+class PackedFitPolicy(AllocationPolicy):
+    """MI355X-aware scorer (the reference left this plug-point as empty stubs,
+    instaslice_controller.go:456-469). Score terms, highest wins:
+
+    1. GPU already in the target compute mode, most-occupied first: packing
+       same-mode GPUs tight minimizes the number of GPUs pinned to a mode,
+       keeping idle GPUs reconfigurable — fragmentation on AMD is *per-mode*,
+       not per-slot.
+    2. Among idle GPUs needing a mode change, prefer one whose memory mode
+       already satisfies the profile (avoids the expensive NPS flip), then
+       lower physical index (keeps high-index GPUs free for large SPX jobs,
+       and co-locates small partitions on few GPUs so xGMI links of the
+       remaining GPUs stay uncontended for multi-GPU tenants).
+    """
+
+    name = "packed-fit"
+
+    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+        best: Optional[Tuple[Tuple, Placement]] = None
+        for v in views:
+            if v.compute_mode is profile.compute:
+                p = _place_on(v, profile, needs_change=False)
+                if p is None:
+                    continue
+                # tier 2 = same mode; prefer most-occupied, then lowest index
+                score = (2, len(v.occupied), -v.index)
+            else:
+                p = _place_on(v, profile, needs_change=True)
+                if p is None:
+                    continue
+                mem_ok = v.memory_mode in VALID_MEMORY_MODES[profile.compute]
+                score = (1, 1 if mem_ok else 0, -v.index)
+            if best is None or score > best[0]:
+                best = (score, p)
+        return best[1] if best else None
+
+
+POLICIES: Dict[str, AllocationPolicy] = {
+    p.name: p for p in (FirstFitPolicy(), PackedFitPolicy())
+}
+
+
+def get_policy(name: str) -> AllocationPolicy:
+    if name not in POLICIES:
+        raise KeyError(f"unknown policy {name!r}; have {sorted(POLICIES)}")
+    return POLICIES[name]

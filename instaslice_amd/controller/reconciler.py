@@ -1,0 +1,265 @@
+"""L1 cluster controller: gated-pod admission, placement, ungating, teardown.
+
+Reference analog: InstasliceReconciler (instaslice_controller.go:64-238) with
+the call stack documented in SURVEY.md §3.1/§3.4. State machine (unchanged):
+
+    pod gated, no allocation   -> place -> Allocations[podUUID] = creating
+    allocation created (agent) -> remove scheduling gate -> status = ungated
+    pod deleting (gated)       -> remove finalizer, status = deleted
+    pod deleting (was running) -> after grace period: finalizer off, deleted
+
+Differences by design (not omissions):
+  - placement is mode-aware (controller/policy.py) instead of MIG-slot
+    first-fit;
+  - all CR mutations go through update_with_retry (conflict-absorbing)
+    instead of a 1 s requeue on conflict;
+  - errors from the placer/store fail the reconcile loudly (the engine
+    backs off and retries) instead of being logged and dropped.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Tuple
+
+from instaslice_amd.api.types import (
+    AllocationDetails,
+    AllocationStatus,
+    pod_is_gated,
+    pod_limits,
+    remove_finalizer,
+    ungate_pod,
+)
+from instaslice_amd.controller.policy import AllocationPolicy, build_gpu_views, get_policy
+from instaslice_amd.partition.profiles import (
+    PartitionProfile,
+    ProfileCatalog,
+    extract_profile_from_limits,
+)
+from instaslice_amd.runtime.engine import Engine, Key, Result, WatchSpec
+from instaslice_amd.store.memstore import MemStore, NotFound
+from instaslice_amd.utils import get_logger
+
+INSTASLICE_NS = "instaslice-system"
+
+# Requeue cadences (reference: instaslice_controller.go:225 no-capacity 5 s,
+# :231 no-node-fits 2 s). Configurable because they bound allocation latency
+# under contention.
+REQUEUE_NO_FIT_S = 0.5
+
+
+class PodController:
+    def __init__(
+        self,
+        store: MemStore,
+        policy: str = "packed-fit",
+        teardown_grace_s: float = 30.0,
+        requeue_no_fit_s: float = REQUEUE_NO_FIT_S,
+    ) -> None:
+        self.store = store
+        self.policy: AllocationPolicy = get_policy(policy)
+        self.teardown_grace_s = teardown_grace_s
+        self.requeue_no_fit_s = requeue_no_fit_s
+        self.log = get_logger("controller")
+        # latency observability: pod_uid -> submit time, and measured p50 feed
+        self.alloc_latency_s: List[float] = []
+        self._pending_since: dict = {}
+        self.engine = Engine(
+            name="controller",
+            store=store,
+            reconcile=self._reconcile,
+            watches=[
+                WatchSpec(kind="Pod"),
+                WatchSpec(kind="Instaslice", map_fn=self._instaslice_to_pods),
+            ],
+        )
+
+    # -- watch mapping ------------------------------------------------------
+    # Reference: podMapFunc maps an Instaslice with "created" allocations back
+    # to its pods (instaslice_controller.go:398-407).
+
+    def _instaslice_to_pods(self, event_type: str, obj: dict) -> List[Key]:
+        keys: List[Key] = []
+        for alloc in (obj.get("spec", {}).get("allocations") or {}).values():
+            if alloc.get("allocationStatus") == AllocationStatus.CREATED:
+                keys.append(("Pod", alloc["namespace"], alloc["podName"]))
+        return keys
+
+    # -- helpers ------------------------------------------------------------
+
+    def _find_allocation(self, pod_uid: str) -> Optional[Tuple[dict, dict]]:
+        """Return (instaslice_cr, allocation_dict) holding this pod's
+        allocation, or None."""
+        for cr in self.store.list("Instaslice"):
+            alloc = (cr.get("spec", {}).get("allocations") or {}).get(pod_uid)
+            if alloc:
+                return cr, alloc
+        return None
+
+    def _profile_for(self, cr: dict, name: str) -> Optional[PartitionProfile]:
+        placements = cr.get("spec", {}).get("placements") or {}
+        if placements:
+            cat = ProfileCatalog.from_dict(placements)
+            p = cat.by_name(name)
+            if p:
+                return p
+        return None
+
+    def _try_place(self, pod: dict, profile_name: str) -> Optional[Result]:
+        """Scan nodes, place, persist the allocation (status=creating).
+        Reference: findDeviceForASlice loop (instaslice_controller.go:192-222)."""
+        md = pod["metadata"]
+        node_selector = pod.get("spec", {}).get("nodeSelector") or {}
+        want_node = node_selector.get("kubernetes.io/hostname")
+        for cr in self.store.list("Instaslice"):
+            node_name = cr["metadata"]["name"]
+            if want_node and node_name != want_node:
+                continue
+            profile = self._profile_for(cr, profile_name)
+            if profile is None:
+                continue  # node does not offer this profile
+            views = build_gpu_views(node_name, cr.get("spec", {}))
+            placement = self.policy.place(profile, views)
+            if placement is None:
+                continue
+            alloc = AllocationDetails(
+                profile=profile_name,
+                gpu_uuid=placement.gpu_uuid,
+                ordinal=placement.ordinal,
+                start=placement.ordinal * profile.xcds,
+                size=profile.xcds,
+                pod_uuid=md["uid"],
+                pod_name=md["name"],
+                namespace=md.get("namespace", "default"),
+                nodename=node_name,
+                allocation_status=AllocationStatus.CREATING,
+                compute_mode=placement.compute_mode,
+                memory_mode=placement.memory_mode,
+            )
+
+            def add_alloc(cr_obj: dict, a=alloc) -> Optional[dict]:
+                spec = cr_obj.setdefault("spec", {})
+                allocs = spec.setdefault("allocations", {})
+                if a.pod_uuid in allocs:
+                    return None  # raced with ourselves; done
+                # re-validate the placement against the fresh CR (another
+                # pod may have taken the ordinal between list() and now)
+                fresh_views = build_gpu_views(cr_obj["metadata"]["name"], spec)
+                for v in fresh_views:
+                    if v.uuid == a.gpu_uuid:
+                        if a.ordinal in v.occupied:
+                            return None  # lost the race; requeue will re-place
+                        if (
+                            v.compute_mode.value != a.compute_mode
+                            and v.occupied
+                        ):
+                            return None
+                allocs[a.pod_uuid] = a.to_dict()
+                return cr_obj
+
+            updated = self.store.update_with_retry(
+                "Instaslice", node_name, INSTASLICE_NS, add_alloc
+            )
+            if updated and md["uid"] in updated["spec"]["allocations"]:
+                self.log.debug(
+                    "placed pod %s profile %s on %s/%s ordinal %d",
+                    md["name"], profile_name, node_name,
+                    placement.gpu_uuid[:8], placement.ordinal,
+                )
+                return Result()  # wait for agent's "created" event
+        return Result(requeue_after=self.requeue_no_fit_s)
+
+    def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str) -> None:
+        def mut(cr: dict) -> Optional[dict]:
+            alloc = (cr.get("spec", {}).get("allocations") or {}).get(pod_uid)
+            if not alloc or alloc["allocationStatus"] == status:
+                return None
+            alloc["allocationStatus"] = status
+            return cr
+
+        self.store.update_with_retry("Instaslice", cr_name, INSTASLICE_NS, mut)
+
+    # -- reconcile ------------------------------------------------------------
+
+    def _reconcile(self, key: Key) -> Result:
+        _, namespace, name = key
+        try:
+            pod = self.store.get("Pod", name, namespace)
+        except NotFound:
+            return Result()
+        md = pod["metadata"]
+        uid = md["uid"]
+
+        # teardown path (reference: instaslice_controller.go:99-142)
+        if md.get("deletionTimestamp"):
+            gated = pod_is_gated(pod)
+            if not gated:
+                elapsed = time.time() - float(md["deletionTimestamp"])
+                if elapsed < self.teardown_grace_s:
+                    return Result(requeue_after=self.teardown_grace_s - elapsed)
+            found = self._find_allocation(uid)
+            if found:
+                cr, alloc = found
+                if alloc["allocationStatus"] != AllocationStatus.DELETED:
+                    self._set_allocation_status(
+                        cr["metadata"]["name"], uid, AllocationStatus.DELETED
+                    )
+            # remove our finalizer; the store drops the pod when none remain
+            def strip(p: dict) -> Optional[dict]:
+                from instaslice_amd import FINALIZER_NAME
+
+                if FINALIZER_NAME not in (p["metadata"].get("finalizers") or []):
+                    return None
+                return remove_finalizer(p)
+
+            self.store.update_with_retry("Pod", name, namespace, strip)
+            return Result()
+
+        if not pod_is_gated(pod):
+            return Result()
+
+        # admission path
+        found = self._find_allocation(uid)
+        if found is None:
+            limits = pod_limits(pod)
+            profile_name = extract_profile_from_limits(limits)
+            if profile_name is None:
+                return Result()  # not our pod
+            if uid not in self._pending_since:
+                self._pending_since[uid] = time.monotonic()
+            return self._try_place(pod, profile_name)
+
+        cr, alloc = found
+        status = alloc["allocationStatus"]
+        if status == AllocationStatus.CREATED:
+            # agent realized the partition: let the pod schedule
+            def do_ungate(p: dict) -> Optional[dict]:
+                new = ungate_pod(p)
+                if new["spec"]["schedulingGates"] == p["spec"].get("schedulingGates"):
+                    return None
+                # gate removal unblocks kube-scheduler; reflect it in status
+                new["status"]["phase"] = "Pending"
+                new["status"]["conditions"] = [
+                    {"type": "PodScheduled", "status": "True", "message": "ungated"}
+                ]
+                return new
+
+            self.store.update_with_retry("Pod", name, namespace, do_ungate)
+            self._set_allocation_status(
+                cr["metadata"]["name"], uid, AllocationStatus.UNGATED
+            )
+            t0 = self._pending_since.pop(uid, None)
+            if t0 is not None:
+                self.alloc_latency_s.append(time.monotonic() - t0)
+            return Result()
+        # creating / ungated / deleted: nothing to do here
+        return Result()
+
+    # -- lifecycle ------------------------------------------------------------
+
+    def start(self) -> "PodController":
+        self.engine.start()
+        return self
+
+    def stop(self) -> None:
+        self.engine.stop()

@@ -1,0 +1,214 @@
+"""In-memory API-server double: typed object store with watch semantics.
+
+The reference's entire coordination bus is the Kubernetes API server — the
+Instaslice CR plus status strings form the controller<->daemonset protocol
+(SURVEY.md §1). This store reproduces the API-server behaviors that protocol
+relies on:
+
+  - optimistic concurrency: every object carries a resourceVersion; an update
+    against a stale version raises Conflict (the reference requeues after 1 s
+    on conflicts, instaslice_controller.go:93)
+  - watches: subscribers receive (event_type, object) for ADDED/MODIFIED/
+    DELETED, driving the event-driven reconcilers (controller-runtime analog)
+  - deletion with finalizers: delete() sets deletionTimestamp if finalizers
+    remain (exactly k8s semantics; the reference depends on this for the
+    two-phase teardown, instaslice_controller.go:99-142)
+
+It is the envtest analog for our integration tests (reference test tier 1,
+internal/controller/suite_test.go:52-84) and the backing of the TCP-served
+store used by multi-rank benchmarks (store/netstore.py).
+"""
+
+from __future__ import annotations
+
+import copy
+import queue
+import threading
+from typing import Callable, Dict, List, Optional, Tuple
+
+
+class Conflict(Exception):
+    """Update with stale resourceVersion (HTTP 409 analog)."""
+
+
+class NotFound(Exception):
+    """Object does not exist (HTTP 404 analog)."""
+
+
+class AlreadyExists(Exception):
+    """Create of an existing object (HTTP 409 analog)."""
+
+
+Key = Tuple[str, str, str]  # (kind, namespace, name)
+
+
+def _key(obj: dict) -> Key:
+    md = obj.get("metadata", {})
+    return (obj["kind"], md.get("namespace", ""), md["name"])
+
+
+class Watch:
+    """A subscription delivering (event_type, object) tuples.
+
+    event_type is "ADDED" | "MODIFIED" | "DELETED". Objects are deep copies.
+    """
+
+    def __init__(self, store: "MemStore", kind: Optional[str]):
+        self._store = store
+        self.kind = kind
+        self._q: "queue.Queue[Optional[Tuple[str, dict]]]" = queue.Queue()
+        self._stopped = False
+
+    def _push(self, event: Tuple[str, dict]) -> None:
+        if not self._stopped:
+            self._q.put(event)
+
+    def next(self, timeout: Optional[float] = None) -> Optional[Tuple[str, dict]]:
+        """Blocking next event; None on stop or timeout."""
+        try:
+            return self._q.get(timeout=timeout)
+        except queue.Empty:
+            return None
+
+    def stop(self) -> None:
+        self._stopped = True
+        self._q.put(None)
+
+
+class MemStore:
+    """Thread-safe object store with k8s-like verbs.
+
+    All returned objects are deep copies: mutating them does not change the
+    store (matching client-go cache semantics the reference relies on).
+    """
+
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        self._objects: Dict[Key, dict] = {}
+        self._rv = 0
+        self._watches: List[Watch] = []
+
+    # -- verbs ------------------------------------------------------------
+
+    def create(self, obj: dict) -> dict:
+        with self._lock:
+            k = _key(obj)
+            if k in self._objects:
+                raise AlreadyExists(f"{k} already exists")
+            obj = copy.deepcopy(obj)
+            self._rv += 1
+            obj.setdefault("metadata", {})["resourceVersion"] = str(self._rv)
+            self._objects[k] = obj
+            self._notify("ADDED", obj)
+            return copy.deepcopy(obj)
+
+    def get(self, kind: str, name: str, namespace: str = "") -> dict:
+        with self._lock:
+            k = (kind, namespace, name)
+            if k not in self._objects:
+                raise NotFound(f"{k} not found")
+            return copy.deepcopy(self._objects[k])
+
+    def list(self, kind: str, namespace: Optional[str] = None) -> List[dict]:
+        with self._lock:
+            return [
+                copy.deepcopy(o)
+                for (knd, ns, _), o in sorted(self._objects.items())
+                if knd == kind and (namespace is None or ns == namespace)
+            ]
+
+    def update(self, obj: dict) -> dict:
+        """Replace; raises Conflict unless obj.resourceVersion matches."""
+        with self._lock:
+            k = _key(obj)
+            if k not in self._objects:
+                raise NotFound(f"{k} not found")
+            cur = self._objects[k]
+            sent_rv = obj.get("metadata", {}).get("resourceVersion")
+            if sent_rv is not None and sent_rv != cur["metadata"]["resourceVersion"]:
+                raise Conflict(
+                    f"{k}: resourceVersion {sent_rv} != {cur['metadata']['resourceVersion']}"
+                )
+            obj = copy.deepcopy(obj)
+            self._rv += 1
+            obj["metadata"]["resourceVersion"] = str(self._rv)
+            # deletionTimestamp is sticky (k8s semantics)
+            if cur["metadata"].get("deletionTimestamp") and not obj["metadata"].get(
+                "deletionTimestamp"
+            ):
+                obj["metadata"]["deletionTimestamp"] = cur["metadata"]["deletionTimestamp"]
+            self._objects[k] = obj
+            # finalizer-free object already marked deleted -> actually remove
+            if obj["metadata"].get("deletionTimestamp") and not obj["metadata"].get(
+                "finalizers"
+            ):
+                del self._objects[k]
+                self._notify("DELETED", obj)
+            else:
+                self._notify("MODIFIED", obj)
+            return copy.deepcopy(obj)
+
+    def delete(self, kind: str, name: str, namespace: str = "", *, now: float = 0.0) -> None:
+        """k8s-style delete: with finalizers present, only sets
+        deletionTimestamp (a MODIFIED event); otherwise removes."""
+        with self._lock:
+            k = (kind, namespace, name)
+            if k not in self._objects:
+                raise NotFound(f"{k} not found")
+            obj = self._objects[k]
+            if obj["metadata"].get("finalizers"):
+                if not obj["metadata"].get("deletionTimestamp"):
+                    import time
+
+                    obj["metadata"]["deletionTimestamp"] = now or time.time()
+                    self._rv += 1
+                    obj["metadata"]["resourceVersion"] = str(self._rv)
+                    self._notify("MODIFIED", obj)
+            else:
+                del self._objects[k]
+                self._notify("DELETED", obj)
+
+    # -- watch ------------------------------------------------------------
+
+    def watch(self, kind: Optional[str] = None, *, replay: bool = True) -> Watch:
+        """Subscribe to events for `kind` (None = all kinds). With replay,
+        current objects are delivered first as ADDED (k8s informer analog)."""
+        with self._lock:
+            w = Watch(self, kind)
+            if replay:
+                for (knd, _, _), o in sorted(self._objects.items()):
+                    if kind is None or knd == kind:
+                        w._push(("ADDED", copy.deepcopy(o)))
+            self._watches.append(w)
+            return w
+
+    def _notify(self, event_type: str, obj: dict) -> None:
+        for w in list(self._watches):
+            if w._stopped:
+                self._watches.remove(w)
+                continue
+            if w.kind is None or w.kind == obj["kind"]:
+                w._push((event_type, copy.deepcopy(obj)))
+
+    # -- helpers ----------------------------------------------------------
+
+    def update_with_retry(
+        self, kind: str, name: str, namespace: str, mutate: Callable[[dict], Optional[dict]],
+        attempts: int = 10,
+    ) -> Optional[dict]:
+        """Get-mutate-update loop absorbing Conflicts. `mutate` returns the
+        modified object or None to abort. This is the pattern the reference
+        implements as requeue-on-conflict (instaslice_controller.go:93)."""
+        for _ in range(attempts):
+            try:
+                obj = self.get(kind, name, namespace)
+            except NotFound:
+                return None
+            new = mutate(obj)
+            if new is None:
+                return None
+            try:
+                return self.update(new)
+            except Conflict:
+                continue
+        raise Conflict(f"update_with_retry: {attempts} attempts exhausted for {kind}/{name}")

@@ -1,0 +1,225 @@
+"""Integration tests: full controller + agent stack over FakeAmdSmi.
+
+This is the tier the reference left broken (its daemonset unit test calls a
+method that doesn't exist) — here it is the backbone:
+the complete creating -> created -> ungated -> deleted machine runs against
+a fake 8x MI355X node with no hardware.
+"""
+
+import pytest
+
+from instaslice_amd.api.types import AllocationStatus
+from instaslice_amd.runtime.cluster import Cluster
+from instaslice_amd.smi import FakeAmdSmi, SmiError
+
+
+@pytest.fixture
+def cluster():
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=8, node_name="node-0"))
+    c.start()
+    yield c
+    c.stop()
+
+
+def test_happy_path_single_pod(cluster):
+    cluster.submit_pod("p1", "cpx-1x36")
+    pod = cluster.wait_pod_scheduled("p1")
+    assert not pod["spec"]["schedulingGates"]
+    env = cluster.pod_env("p1")
+    assert "ROCR_VISIBLE_DEVICES" in env and env["ROCR_VISIBLE_DEVICES"]
+    assert env["INSTASLICE_PARTITION_ORDINAL"] == "0"
+    allocs = cluster.allocations("node-0")
+    (alloc,) = allocs.values()
+    assert alloc["allocationStatus"] == AllocationStatus.UNGATED
+    assert alloc["profile"] == "cpx-1x36"
+    assert alloc["computeMode"] == "CPX"
+    prepared = cluster.prepared("node-0")
+    assert len(prepared) == 1
+    (prep,) = prepared.values()
+    assert prep["xcds"] == 1 and prep["memoryGB"] == 36
+    # node capacity pinning (reference: org.instaslice/<pod> extended resource)
+    node = cluster.store.get("Node", "node-0", "")
+    assert node["status"]["capacity"].get("org.instaslice/p1") == 1
+
+
+def test_gpu_actually_reconfigured(cluster):
+    cluster.submit_pod("p1", "qpx-2x72")
+    cluster.wait_pod_scheduled("p1")
+    smi = cluster.agents["node-0"].smi
+    gpus = smi.list_gpus()
+    modes = {g.compute_mode for g in gpus}
+    assert "QPX" in modes
+    # reconfigure telemetry captured (north-star requirement)
+    events = cluster.agents["node-0"].reconfigure_events
+    assert len(events) == 1
+    assert events[0]["to"].startswith("QPX")
+    assert events[0]["set_wall_ms"] >= 0
+
+
+def test_eight_cpx_pods_fill_one_gpu(cluster):
+    for i in range(8):
+        cluster.submit_pod(f"p{i}", "cpx-1x36")
+    for i in range(8):
+        cluster.wait_pod_scheduled(f"p{i}")
+    # packed-fit: all 8 pods share one CPX GPU; only one mode change happened
+    prepared = cluster.prepared("node-0")
+    assert len(prepared) == 8
+    parents = {p["parentGpuUUID"] for p in prepared.values()}
+    assert len(parents) == 1
+    ordinals = sorted(p["ordinal"] for p in prepared.values())
+    assert ordinals == list(range(8))
+    assert len(cluster.agents["node-0"].reconfigure_events) == 1
+    # distinct visible devices per pod
+    uuids = {cluster.pod_env(f"p{i}")["ROCR_VISIBLE_DEVICES"] for i in range(8)}
+    assert len(uuids) == 8
+
+
+def test_mixed_profiles_across_gpus(cluster):
+    cluster.submit_pod("spx", "spx-8x288")
+    cluster.submit_pod("dpx-a", "dpx-4x144")
+    cluster.submit_pod("dpx-b", "dpx-4x144")
+    for name in ("spx", "dpx-a", "dpx-b"):
+        cluster.wait_pod_scheduled(name)
+    prepared = cluster.prepared("node-0")
+    assert len(prepared) == 3
+    # the two DPX pods share one GPU; SPX takes a whole other one
+    by_parent = {}
+    for p in prepared.values():
+        by_parent.setdefault(p["parentGpuUUID"], []).append(p)
+    sizes = sorted(len(v) for v in by_parent.values())
+    assert sizes == [1, 2]
+
+
+def test_delete_cycle_cleans_everything(cluster):
+    cluster.submit_pod("p1", "cpx-1x36")
+    cluster.wait_pod_scheduled("p1")
+    cluster.delete_pod("p1")
+    cluster.wait_pod_gone("p1")
+    cluster.wait_allocations_empty("node-0")
+    from instaslice_amd.store import NotFound
+    with pytest.raises(NotFound):
+        cluster.store.get("ConfigMap", "p1", "default")
+    node = cluster.store.get("Node", "node-0", "")
+    assert "org.instaslice/p1" not in node["status"]["capacity"]
+
+
+def test_slot_reused_after_delete(cluster):
+    cluster.submit_pod("p1", "spx-8x288")
+    cluster.wait_pod_scheduled("p1")
+    # 8 GPUs, all SPX: second + ... + ninth pods fit, tenth must wait
+    for i in range(2, 9):
+        cluster.submit_pod(f"p{i}", "spx-8x288")
+        cluster.wait_pod_scheduled(f"p{i}")
+    cluster.submit_pod("p9", "spx-8x288")
+    import time
+    time.sleep(0.2)
+    pod = cluster.store.get("Pod", "p9", "default")
+    assert pod["spec"]["schedulingGates"], "9th SPX pod must be gated (no capacity)"
+    # free one -> the waiter lands
+    cluster.delete_pod("p1")
+    cluster.wait_pod_gone("p1")
+    cluster.wait_pod_scheduled("p9", timeout=15.0)
+
+
+def test_gated_pod_deleted_before_allocation_runs(cluster):
+    """Reference path: gated pod deleted before it ran -> finalizer removed,
+    no residue (instaslice_controller.go:89-98)."""
+    cluster.submit_pod("p1", "cpx-1x36")
+    cluster.wait_pod_scheduled("p1")
+    cluster.delete_pod("p1")
+    cluster.wait_pod_gone("p1")
+    cluster.wait_allocations_empty("node-0")
+
+
+def test_mode_sticky_after_drain(cluster):
+    """Default policy: a drained GPU keeps its mode so the next same-profile
+    pod needs no reconfiguration (AMD-native optimization; reference destroys
+    slices on teardown)."""
+    cluster.submit_pod("p1", "cpx-1x36")
+    cluster.wait_pod_scheduled("p1")
+    cluster.delete_pod("p1")
+    cluster.wait_pod_gone("p1")
+    cluster.wait_allocations_empty("node-0")
+    n_events = len(cluster.agents["node-0"].reconfigure_events)
+    cluster.submit_pod("p2", "cpx-1x36")
+    cluster.wait_pod_scheduled("p2")
+    assert len(cluster.agents["node-0"].reconfigure_events) == n_events
+
+
+def test_smi_failure_fails_allocation_loudly():
+    """The reference logs-and-ignores NVML errors (instaslice_daemonset.go:173-189);
+    we require the error to surface and the pod to stay gated."""
+    c = Cluster(teardown_grace_s=0.0)
+    smi = FakeAmdSmi(num_gpus=1, node_name="node-0")
+
+    def hook(verb, gpu):
+        if verb == "set_compute_partition":
+            raise SmiError("injected driver failure")
+
+    c.add_node("node-0", smi)
+    c.start()
+    try:
+        smi.fault_hook = hook
+        c.submit_pod("p1", "cpx-1x36")
+        with pytest.raises(TimeoutError):
+            c.wait_pod_scheduled("p1", timeout=0.5)
+        assert c.agents["node-0"].engine.error_count > 0
+        # heal the fault: the engine retries and the pod schedules
+        smi.fault_hook = None
+        c.wait_pod_scheduled("p1", timeout=10.0)
+    finally:
+        c.stop()
+
+
+def test_two_node_cluster_spillover():
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.add_node("node-1", FakeAmdSmi(num_gpus=1, node_name="node-1"))
+    c.start()
+    try:
+        c.submit_pod("a", "spx-8x288")
+        c.wait_pod_scheduled("a")
+        c.submit_pod("b", "spx-8x288")
+        c.wait_pod_scheduled("b")
+        nodes = {
+            next(iter(c.allocations(n).values()))["nodename"]
+            for n in ("node-0", "node-1")
+            if c.allocations(n)
+        }
+        assert nodes == {"node-0", "node-1"}
+    finally:
+        c.stop()
+
+
+def test_node_selector_pins_pod():
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.add_node("node-1", FakeAmdSmi(num_gpus=1, node_name="node-1"))
+    c.start()
+    try:
+        c.submit_pod("a", "cpx-1x36", node="node-1")
+        c.wait_pod_scheduled("a")
+        assert len(c.allocations("node-1")) == 1
+        assert not c.allocations("node-0")
+    finally:
+        c.stop()
+
+
+def test_agent_restart_adopts_state(cluster):
+    """Dangling-partition adoption (reference: discoverDanglingSlices,
+    instaslice_daemonset.go:666-748): stop the agent, restart a new one over
+    the same store + smi; prepared state survives and new pods still place."""
+    cluster.submit_pod("p1", "cpx-1x36")
+    cluster.wait_pod_scheduled("p1")
+    old = cluster.agents["node-0"]
+    smi = old.smi
+    old.stop()
+    new = cluster.add_node("node-0", smi)
+    new.start()
+    assert len(cluster.prepared("node-0")) == 1
+    cluster.submit_pod("p2", "cpx-1x36")
+    cluster.wait_pod_scheduled("p2")
+    ords = sorted(p["ordinal"] for p in cluster.prepared("node-0").values())
+    assert ords == [0, 1]
+    new.stop()

@@ -1,0 +1,143 @@
+"""Placer tests: first-fit + packed-fit over the MI355X mode matrix.
+
+Exhaustive over profile mixes per SURVEY.md §7.2 item 6 ('unit: placer
+exhaustive over profile mixes').
+"""
+
+import itertools
+
+import pytest
+
+from instaslice_amd.controller.policy import (
+    FirstFitPolicy,
+    GpuView,
+    PackedFitPolicy,
+    Placement,
+)
+from instaslice_amd.partition.profiles import ComputeMode, MemoryMode, mi355x_catalog
+
+CAT = mi355x_catalog()
+CPX = CAT.by_name("cpx-1x36")
+QPX = CAT.by_name("qpx-2x72")
+DPX = CAT.by_name("dpx-4x144")
+SPX = CAT.by_name("spx-8x288")
+
+
+def mk_view(idx, mode=ComputeMode.SPX, mem=MemoryMode.NPS1, occupied=()):
+    return GpuView(
+        node="n0", uuid=f"gpu-{idx}", index=idx, memory_gb=288,
+        compute_mode=mode, memory_mode=mem, occupied=set(occupied),
+    )
+
+
+def test_firstfit_prefers_existing_mode():
+    views = [
+        mk_view(0, ComputeMode.SPX),
+        mk_view(1, ComputeMode.CPX, MemoryMode.NPS4, occupied={0, 1}),
+    ]
+    p = FirstFitPolicy().place(CPX, views)
+    assert p.gpu_uuid == "gpu-1" and p.ordinal == 2 and not p.needs_mode_change
+
+
+def test_firstfit_reconfigures_idle_gpu():
+    views = [mk_view(0, ComputeMode.SPX), mk_view(1, ComputeMode.SPX)]
+    p = FirstFitPolicy().place(CPX, views)
+    assert p.gpu_uuid == "gpu-0" and p.ordinal == 0 and p.needs_mode_change
+    assert p.compute_mode == "CPX" and p.memory_mode == "NPS1"  # sticky memory
+
+
+def test_mode_change_requires_idle_gpu():
+    # gpu-0 in SPX but occupied (the SPX partition is allocated): cannot flip
+    views = [mk_view(0, ComputeMode.SPX, occupied={0})]
+    assert FirstFitPolicy().place(CPX, views) is None
+    assert PackedFitPolicy().place(CPX, views) is None
+
+
+def test_full_cpx_gpu_rejects():
+    views = [mk_view(0, ComputeMode.CPX, occupied=set(range(8)))]
+    assert FirstFitPolicy().place(CPX, views) is None
+
+
+def test_memory_mode_planned_when_illegal_under_target():
+    # NPS4 GPU asked to go SPX: NPS4 is illegal under SPX -> plan NPS1
+    views = [mk_view(0, ComputeMode.CPX, MemoryMode.NPS4)]
+    p = FirstFitPolicy().place(SPX, views)
+    assert p.needs_mode_change and p.memory_mode == "NPS1"
+
+
+def test_packedfit_packs_tightest_same_mode_gpu():
+    views = [
+        mk_view(0, ComputeMode.CPX, occupied={0}),
+        mk_view(1, ComputeMode.CPX, occupied={0, 1, 2}),
+        mk_view(2, ComputeMode.SPX),
+    ]
+    p = PackedFitPolicy().place(CPX, views)
+    assert p.gpu_uuid == "gpu-1" and p.ordinal == 3  # most-occupied wins
+
+
+def test_packedfit_prefers_matching_memory_mode_on_reconfig():
+    views = [
+        mk_view(0, ComputeMode.SPX, MemoryMode.NPS1),
+        mk_view(1, ComputeMode.DPX, MemoryMode.NPS1),
+    ]
+    # both idle, both need a change; equal memory validity -> lower index
+    p = PackedFitPolicy().place(CPX, views)
+    assert p.gpu_uuid == "gpu-0"
+
+
+@pytest.mark.parametrize("policy_cls", [FirstFitPolicy, PackedFitPolicy])
+def test_exhaustive_mix_8gpu_node(policy_cls):
+    """Feed every 3-profile request mix to an 8-GPU node, applying each
+    placement; assert invariants: ordinals unique per GPU, mode changes only
+    on idle GPUs, capacity respected."""
+    profiles = [CPX, QPX, DPX, SPX]
+    for mix in itertools.product(profiles, repeat=3):
+        views = [mk_view(i) for i in range(8)]
+        policy = policy_cls()
+        for prof in mix:
+            p = policy.place(prof, views)
+            assert p is not None, f"mix {[m.name for m in mix]} failed to place"
+            v = next(v for v in views if v.uuid == p.gpu_uuid)
+            if p.needs_mode_change:
+                assert not v.occupied
+                v.compute_mode = prof.compute
+                v.memory_mode = MemoryMode(p.memory_mode)
+            assert v.compute_mode is prof.compute
+            assert p.ordinal not in v.occupied
+            assert p.ordinal < prof.partitions_per_gpu
+            v.occupied.add(p.ordinal)
+
+
+@pytest.mark.parametrize("policy_cls", [FirstFitPolicy, PackedFitPolicy])
+def test_capacity_64_cpx_partitions_per_node(policy_cls):
+    """8 GPUs x CPX = 64 schedulable partitions (SURVEY.md §5 'long-context'
+    analog note); the 65th request must not fit."""
+    views = [mk_view(i) for i in range(8)]
+    policy = policy_cls()
+    placed = []
+    for _ in range(64):
+        p = policy.place(CPX, views)
+        assert p is not None
+        v = next(v for v in views if v.uuid == p.gpu_uuid)
+        if p.needs_mode_change:
+            v.compute_mode = ComputeMode.CPX
+        v.occupied.add(p.ordinal)
+        placed.append(p)
+    assert policy.place(CPX, views) is None
+    assert len({(p.gpu_uuid, p.ordinal) for p in placed}) == 64
+
+
+def test_packedfit_leaves_room_for_spx():
+    """Packing keeps whole GPUs free: 7 CPX pods should land on one GPU so an
+    SPX (whole-GPU) job still fits on the other."""
+    views = [mk_view(0), mk_view(1)]
+    policy = PackedFitPolicy()
+    for _ in range(7):
+        p = policy.place(CPX, views)
+        v = next(v for v in views if v.uuid == p.gpu_uuid)
+        if p.needs_mode_change:
+            v.compute_mode = ComputeMode.CPX
+        v.occupied.add(p.ordinal)
+    assert views[0].compute_mode is ComputeMode.CPX and len(views[0].occupied) == 7
+    p = policy.place(SPX, views)
+    assert p is not None and p.gpu_uuid == "gpu-1"

@@ -1,0 +1,351 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: dynamic GPU-partition allocation throughput on MI355X.
+
+Measures the BASELINE.json north-star metric — pods scheduled per second and
+p50 slice-allocation latency — on synthetic sleep pods with random partition
+profile requests, against real amd-smi partitioning when a GPU is present
+(FakeAmdSmi otherwise, recorded in config.backend).
+
+One step = one full pod lifecycle on this rank's node:
+  submit gated pod -> controller places it -> node agent realizes the
+  partition (whole-GPU mode set via libamd_smi when needed) -> pod ungated
+  (latency sample) -> HIP payload kernel runs inside the allocated partition
+  -> pod deleted -> allocation drained.
+
+Topology: rank 0 hosts the store (TCP) + the cluster controller; every rank
+(including 0) runs one node agent managing its own GPU. Launch:
+
+  python bench.py --gpus 1 --steps 100 --warmup 10            # single GPU
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ...            # N GPUs
+
+Weak scaling: per-GPU work is fixed (each rank drives K pods at its node);
+value = total pods / max-over-ranks elapsed seconds.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import statistics
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+from instaslice_amd.agent.daemonset import NodeAgent
+from instaslice_amd.controller.reconciler import PodController
+from instaslice_amd.partition.profiles import ProfileCatalog
+from instaslice_amd.runtime.cluster import Cluster  # noqa: F401 (doc anchor)
+from instaslice_amd.smi.base import AmdSmi, SmiBusy, SmiError
+from instaslice_amd.smi.fake import FakeAmdSmi
+from instaslice_amd.store.memstore import MemStore, NotFound
+from instaslice_amd.store.netstore import NetStoreClient, StoreServer
+from instaslice_amd.utils import get_logger
+
+log = get_logger("bench")
+
+
+class SingleGpuSmi(AmdSmi):
+    """Restrict a node-wide SMI to one physical GPU (rank -> GPU binding)."""
+
+    def __init__(self, inner: AmdSmi, gpu_index: int):
+        self.inner = inner
+        self.gpu_index = gpu_index
+
+    def init(self):
+        self.inner.init()
+
+    def shutdown(self):
+        self.inner.shutdown()
+
+    def list_gpus(self):
+        gpus = self.inner.list_gpus()
+        mine = [g for g in gpus if g.index == self.gpu_index]
+        for g in mine:
+            g.index = 0
+        return mine
+
+    def get_compute_partition(self, u):
+        return self.inner.get_compute_partition(u)
+
+    def set_compute_partition(self, u, m):
+        self.inner.set_compute_partition(u, m)
+
+    def get_memory_partition(self, u):
+        return self.inner.get_memory_partition(u)
+
+    def set_memory_partition(self, u, m):
+        self.inner.set_memory_partition(u, m)
+
+    def get_profile_config(self, u):
+        return self.inner.get_profile_config(u)
+
+    def get_metrics(self, u):
+        return self.inner.get_metrics(u)
+
+
+def make_smi(args, rank: int):
+    """Real amd-smi if a GPU is visible, else the fake 1x MI355X model."""
+    if not args.fake and torch.cuda.is_available():
+        try:
+            from instaslice_amd.smi.native import NativeAmdSmi
+
+            smi = NativeAmdSmi()
+            smi.init()
+            gpus = smi.list_gpus()
+            if gpus:
+                local = int(os.environ.get("LOCAL_RANK", rank))
+                return SingleGpuSmi(smi, min(local, len(gpus) - 1)), "amdsmi"
+            log.warning("amdsmi enumerated no GPUs; falling back to fake")
+        except SmiError as e:
+            log.warning("native smi unavailable (%s); falling back to fake", e)
+    return FakeAmdSmi(num_gpus=1, node_name=f"node-{rank}"), "fake"
+
+
+def probe_partitioning(smi: AmdSmi) -> bool:
+    """Can this box actually set compute partitions? (VM guests often can't.)
+    Probe by re-setting the current mode — a no-op when supported."""
+    try:
+        g = smi.list_gpus()[0]
+        smi.set_compute_partition(g.uuid, g.compute_mode)
+        return True
+    except SmiBusy:
+        return True  # refused for business, not capability
+    except SmiError as e:
+        log.warning("compute-partition set unsupported here: %s", e)
+        return False
+
+
+def choose_profiles(store, node: str, live_partitioning: bool, mix: str, rank: int):
+    """Resolve the per-step profile sequence from the node's discovered
+    catalog. Random cpx/qpx mix when partitioning is live (BASELINE.json
+    'random slice-profile requests'), else the current-mode profile."""
+    deadline = time.monotonic() + 30
+    cr = None
+    while time.monotonic() < deadline:
+        try:
+            cr = store.get("Instaslice", node, "instaslice-system")
+            break
+        except NotFound:
+            time.sleep(0.05)
+    if cr is None:
+        raise RuntimeError(f"node {node} never published its Instaslice CR")
+    cat = ProfileCatalog.from_dict(cr["spec"]["placements"])
+    gpus = cr["spec"]["gpus"]
+    current_mode = next(iter(gpus.values()))["computeMode"]
+
+    if mix == "current" or not live_partitioning:
+        prof = next(p for p in cat.profiles if p.compute.value == current_mode)
+        return [prof.name], "static-" + current_mode.lower()
+    if mix == "cpx":
+        prof = next(p for p in cat.profiles if p.compute.value == "CPX")
+        return [prof.name], "live"
+    # random: mix of CPX and QPX shapes (mode flips exercised)
+    names = [p.name for p in cat.profiles if p.compute.value in ("CPX", "QPX")]
+    if not names:
+        names = [cat.profiles[0].name]
+    return names, "live"
+
+
+def run_payload(device_index: int, enabled: bool) -> None:
+    if not enabled:
+        return
+    from instaslice_amd.ops import _payload
+
+    err = _payload.run_vecadd(1 << 18, device=device_index)
+    if err != 0.0:
+        raise RuntimeError(f"payload vecadd FAILED in partition: max_err={err}")
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=100)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--profile-mix", choices=["random", "cpx", "current"],
+                    default="random")
+    ap.add_argument("--policy", default="packed-fit")
+    ap.add_argument("--fake", action="store_true",
+                    help="force FakeAmdSmi even if a GPU is present")
+    ap.add_argument("--no-payload", action="store_true",
+                    help="skip the HIP kernel run inside each partition")
+    ap.add_argument("--seed", type=int, default=1234)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    has_gpu = torch.cuda.is_available()
+    payload_on = has_gpu and not args.fake and not args.no_payload
+
+    if world > 1:
+        backend = "nccl" if has_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+        if has_gpu:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+
+    # -- topology: rank 0 = store server + controller; all ranks = 1 agent --
+    server = controller = None
+    if rank == 0:
+        server = StoreServer().start()
+        store0 = server.store
+        controller = PodController(store0, policy=args.policy, teardown_grace_s=0.0)
+        controller.requeue_no_fit_s = 0.05
+        controller.start()
+        addr = ("127.0.0.1", server.port)
+    else:
+        addr = None
+    if world > 1:
+        holder = [addr]
+        dist.broadcast_object_list(holder, src=0)
+        addr = holder[0]
+    # Every rank — including 0 — drives its agent and step loop through the
+    # TCP store client, so per-GPU cost is identical across ranks and the
+    # driver's weak-scaling efficiency reflects real behavior (only the
+    # controller is co-located with the store, as it would be in production).
+    store = NetStoreClient(addr[0], addr[1])
+
+    node = f"node-{rank}"
+    smi, backend_name = make_smi(args, rank)
+    live = probe_partitioning(smi) if backend_name == "amdsmi" else True
+    agent = NodeAgent(store, smi, node)
+    agent.start()
+
+    profiles, partitioning = choose_profiles(store, node, live, args.profile_mix, rank)
+    rng = random.Random(args.seed + rank)
+    log.info("rank %d: backend=%s partitioning=%s profiles=%s payload=%s",
+             rank, backend_name, partitioning, profiles, payload_on)
+
+    def sync():
+        if has_gpu:
+            torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
+
+    # -- one step = one pod lifecycle, waits driven by watch events ----------
+    latencies_ms = []
+    events = store.watch(None, replay=False)  # all kinds this rank touches
+
+    def wait_event(pred, what: str, timeout: float = 120.0) -> None:
+        deadline = time.monotonic() + timeout
+        while True:
+            remaining = deadline - time.monotonic()
+            if remaining <= 0:
+                raise TimeoutError(f"timed out waiting for {what}")
+            ev = events.next(timeout=min(remaining, 0.5))
+            if ev is None:
+                continue
+            if pred(ev[0], ev[1]):
+                return
+
+    def step(i: int, record: bool) -> None:
+        name = f"bench-r{rank}-{i}"
+        prof = rng.choice(profiles)
+        from instaslice_amd.api.types import new_pod
+
+        t0 = time.perf_counter()
+        store.create(new_pod(name, profile=prof,
+                             node_selector={"kubernetes.io/hostname": node}))
+
+        def scheduled(et, obj):
+            return (obj["kind"] == "Pod"
+                    and obj["metadata"]["name"] == name
+                    and not obj["spec"].get("schedulingGates"))
+
+        wait_event(scheduled, f"pod {name} ({prof}) scheduled")
+        if record:
+            latencies_ms.append((time.perf_counter() - t0) * 1000.0)
+        cm = store.get("ConfigMap", name, "default")
+        run_payload(int(cm["data"]["HIP_VISIBLE_DEVICES"]), payload_on)
+        store.delete("Pod", name, "default")
+
+        state = {"pod_gone": False, "alloc_gone": False}
+
+        def drained(et, obj):
+            if obj["kind"] == "Pod" and obj["metadata"]["name"] == name:
+                if et == "DELETED":
+                    state["pod_gone"] = True
+            elif obj["kind"] == "Instaslice" and obj["metadata"]["name"] == node:
+                allocs = obj["spec"].get("allocations") or {}
+                state["alloc_gone"] = not any(
+                    a["podName"] == name for a in allocs.values()
+                )
+            return state["pod_gone"] and state["alloc_gone"]
+
+        wait_event(drained, f"pod {name} drained")
+
+    for i in range(args.warmup):
+        step(i, record=False)
+
+    sync()
+    t_start = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i, record=True)
+    sync()
+    elapsed = time.perf_counter() - t_start
+
+    # max over ranks (the contract), pooled latency samples
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        pooled = [None] * world
+        dist.all_gather_object(pooled, latencies_ms)
+        all_lat = [x for sub in pooled for x in sub]
+    else:
+        all_lat = latencies_ms
+
+    n_gpus = world if world > 1 else args.gpus
+    if rank == 0:
+        total_pods = args.steps * world
+        value = total_pods / elapsed
+        p50 = statistics.median(all_lat) if all_lat else None
+        p99 = (statistics.quantiles(all_lat, n=100)[98]
+               if len(all_lat) >= 100 else (max(all_lat) if all_lat else None))
+        result = {
+            "metric": "pods_scheduled_per_s",
+            "value": round(value, 3),
+            "unit": "pods/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "fp32",
+            "data": "synthetic",
+            "p50_alloc_latency_ms": round(p50, 3) if p50 is not None else None,
+            "p99_alloc_latency_ms": round(p99, 3) if p99 is not None else None,
+            "config": {
+                "model": "sleep-pod-partition-allocation",
+                "global_batch": total_pods,
+                "seq_len": None,
+                "parallelism": f"1-controller+{world}-node-agents",
+                "profile_mix": profiles,
+                "partitioning": partitioning,
+                "backend": backend_name,
+                "payload_kernel": payload_on,
+                "policy": args.policy,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    # teardown
+    events.stop()
+    agent.stop()
+    if controller:
+        controller.stop()
+    store.close()
+    if server:
+        server.stop()
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

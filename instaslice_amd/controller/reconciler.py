@@ -87,10 +87,13 @@ class PodController:
 
     # -- helpers ------------------------------------------------------------
 
-    def _find_allocation(self, pod_uid: str) -> Optional[Tuple[dict, dict]]:
+    def _find_allocation(
+        self, pod_uid: str, crs: List[dict]
+    ) -> Optional[Tuple[dict, dict]]:
         """Return (instaslice_cr, allocation_dict) holding this pod's
-        allocation, or None."""
-        for cr in self.store.list("Instaslice"):
+        allocation, or None. `crs` is the reconcile's single Instaslice list
+        (one store round-trip per reconcile, shared with placement)."""
+        for cr in crs:
             alloc = (cr.get("spec", {}).get("allocations") or {}).get(pod_uid)
             if alloc:
                 return cr, alloc
@@ -105,13 +108,13 @@ class PodController:
                 return p
         return None
 
-    def _try_place(self, pod: dict, profile_name: str) -> Optional[Result]:
+    def _try_place(self, pod: dict, profile_name: str, crs: List[dict]) -> Optional[Result]:
         """Scan nodes, place, persist the allocation (status=creating).
         Reference: findDeviceForASlice loop (instaslice_controller.go:192-222)."""
         md = pod["metadata"]
         node_selector = pod.get("spec", {}).get("nodeSelector") or {}
         want_node = node_selector.get("kubernetes.io/hostname")
-        for cr in self.store.list("Instaslice"):
+        for cr in crs:
             node_name = cr["metadata"]["name"]
             if want_node and node_name != want_node:
                 continue
@@ -197,7 +200,7 @@ class PodController:
                 elapsed = time.time() - float(md["deletionTimestamp"])
                 if elapsed < self.teardown_grace_s:
                     return Result(requeue_after=self.teardown_grace_s - elapsed)
-            found = self._find_allocation(uid)
+            found = self._find_allocation(uid, self.store.list("Instaslice"))
             if found:
                 cr, alloc = found
                 if alloc["allocationStatus"] != AllocationStatus.DELETED:
@@ -219,7 +222,8 @@ class PodController:
             return Result()
 
         # admission path
-        found = self._find_allocation(uid)
+        crs = self.store.list("Instaslice")
+        found = self._find_allocation(uid, crs)
         if found is None:
             limits = pod_limits(pod)
             profile_name = extract_profile_from_limits(limits)
@@ -227,7 +231,7 @@ class PodController:
                 return Result()  # not our pod
             if uid not in self._pending_since:
                 self._pending_since[uid] = time.monotonic()
-            return self._try_place(pod, profile_name)
+            return self._try_place(pod, profile_name, crs)
 
         cr, alloc = found
         status = alloc["allocationStatus"]

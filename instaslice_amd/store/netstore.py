@@ -1,0 +1,279 @@
+"""TCP-served store: MemStore behind a JSON-lines protocol.
+
+The reference's coordination bus is the Kubernetes API server; ours is this
+store. In-process callers use MemStore directly; multi-process deployments
+(the N-rank benchmark: one controller, N node agents, one per GPU) talk to a
+StoreServer over TCP with the same verb set, including watches.
+
+Protocol (one JSON object per line):
+  request:   {"id": 1, "verb": "get", "kind": "Pod", "name": ..., "namespace": ...}
+  response:  {"id": 1, "ok": true, "result": {...}}
+             {"id": 1, "ok": false, "error": {"type": "NotFound", "msg": ...}}
+  watch:     request {"id": 2, "verb": "watch", "kind": "Pod"} -> ok with
+             {"watch_id": N}; events then stream as
+             {"watch_id": N, "event": ["ADDED", {...}]}
+
+Exceptions round-trip by name so Conflict/NotFound semantics (and therefore
+update_with_retry) behave identically to the in-memory store.
+"""
+
+from __future__ import annotations
+
+import json
+import queue
+import socket
+import socketserver
+import threading
+from typing import Callable, List, Optional, Tuple
+
+from instaslice_amd.store.memstore import (
+    AlreadyExists,
+    Conflict,
+    MemStore,
+    NotFound,
+)
+from instaslice_amd.utils import get_logger
+
+_EXC_BY_NAME = {
+    "Conflict": Conflict,
+    "NotFound": NotFound,
+    "AlreadyExists": AlreadyExists,
+}
+
+
+class _Handler(socketserver.StreamRequestHandler):
+    # small request/reply + async watch pushes on one socket: Nagle + delayed
+    # ACK would add 40 ms stalls (observed as p99 spikes) — disable it
+    disable_nagle_algorithm = True
+
+    def handle(self) -> None:
+        store: MemStore = self.server.store  # type: ignore[attr-defined]
+        log = self.server.log  # type: ignore[attr-defined]
+        wlock = threading.Lock()
+        watches = []
+
+        def send(obj: dict) -> None:
+            data = (json.dumps(obj) + "\n").encode()
+            with wlock:
+                self.wfile.write(data)
+                self.wfile.flush()
+
+        def pump_watch(watch_id: int, w) -> None:
+            while True:
+                ev = w.next(timeout=0.5)
+                if ev is None:
+                    if w._stopped or self.server.stopping:  # type: ignore[attr-defined]
+                        return
+                    continue
+                try:
+                    send({"watch_id": watch_id, "event": list(ev)})
+                except (BrokenPipeError, OSError):
+                    w.stop()
+                    return
+
+        next_watch_id = 0
+        try:
+            for line in self.rfile:
+                if not line.strip():
+                    continue
+                req = json.loads(line)
+                rid = req.get("id")
+                verb = req.get("verb")
+                try:
+                    if verb == "create":
+                        res = store.create(req["obj"])
+                    elif verb == "get":
+                        res = store.get(req["kind"], req["name"], req.get("namespace", ""))
+                    elif verb == "list":
+                        res = store.list(req["kind"], req.get("namespace"))
+                    elif verb == "update":
+                        res = store.update(req["obj"])
+                    elif verb == "delete":
+                        store.delete(req["kind"], req["name"], req.get("namespace", ""))
+                        res = None
+                    elif verb == "watch":
+                        w = store.watch(req.get("kind"), replay=req.get("replay", True))
+                        watches.append(w)
+                        next_watch_id += 1
+                        wid = next_watch_id
+                        threading.Thread(
+                            target=pump_watch, args=(wid, w), daemon=True
+                        ).start()
+                        res = {"watch_id": wid}
+                    elif verb == "ping":
+                        res = "pong"
+                    else:
+                        raise ValueError(f"unknown verb {verb!r}")
+                    send({"id": rid, "ok": True, "result": res})
+                except (Conflict, NotFound, AlreadyExists) as e:
+                    send({"id": rid, "ok": False,
+                          "error": {"type": type(e).__name__, "msg": str(e)}})
+                except Exception as e:  # malformed request: report, keep serving
+                    log.warning("request error: %s", e)
+                    send({"id": rid, "ok": False,
+                          "error": {"type": "Error", "msg": str(e)}})
+        except (ConnectionResetError, BrokenPipeError, OSError):
+            pass
+        finally:
+            for w in watches:
+                w.stop()
+
+
+class StoreServer:
+    """Serve a MemStore on host:port. port=0 picks a free port."""
+
+    def __init__(self, store: Optional[MemStore] = None,
+                 host: str = "127.0.0.1", port: int = 0) -> None:
+        self.store = store or MemStore()
+        self.log = get_logger("netstore.server")
+        self._srv = socketserver.ThreadingTCPServer((host, port), _Handler,
+                                                    bind_and_activate=True)
+        self._srv.daemon_threads = True
+        self._srv.allow_reuse_address = True
+        self._srv.store = self.store  # type: ignore[attr-defined]
+        self._srv.log = self.log  # type: ignore[attr-defined]
+        self._srv.stopping = False  # type: ignore[attr-defined]
+        self.host, self.port = self._srv.server_address
+        self._thread = threading.Thread(target=self._srv.serve_forever,
+                                        name="netstore-server", daemon=True)
+
+    def start(self) -> "StoreServer":
+        self._thread.start()
+        return self
+
+    def stop(self) -> None:
+        self._srv.stopping = True  # type: ignore[attr-defined]
+        self._srv.shutdown()
+        self._srv.server_close()
+
+
+class _ClientWatch:
+    """Client-side watch mirroring memstore.Watch."""
+
+    def __init__(self) -> None:
+        self._q: "queue.Queue[Optional[Tuple[str, dict]]]" = queue.Queue()
+        self._stopped = False
+        self.kind: Optional[str] = None
+
+    def next(self, timeout: Optional[float] = None) -> Optional[Tuple[str, dict]]:
+        try:
+            return self._q.get(timeout=timeout)
+        except queue.Empty:
+            return None
+
+    def stop(self) -> None:
+        self._stopped = True
+        self._q.put(None)
+
+
+class NetStoreClient:
+    """Store client with the MemStore interface (duck-typed)."""
+
+    def __init__(self, host: str, port: int, timeout: float = 30.0) -> None:
+        self._sock = socket.create_connection((host, port), timeout=timeout)
+        self._sock.settimeout(None)
+        self._sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        self._rfile = self._sock.makefile("r")
+        self._wlock = threading.Lock()
+        self._pending: dict = {}
+        self._watches: dict = {}
+        self._next_id = 0
+        self._idlock = threading.Lock()
+        self._closed = False
+        self.log = get_logger("netstore.client")
+        self._reader = threading.Thread(target=self._read_loop,
+                                        name="netstore-reader", daemon=True)
+        self._reader.start()
+
+    def _read_loop(self) -> None:
+        try:
+            for line in self._rfile:
+                msg = json.loads(line)
+                if "watch_id" in msg and "event" in msg:
+                    w = self._watches.get(msg["watch_id"])
+                    if w and not w._stopped:
+                        et, obj = msg["event"]
+                        w._q.put((et, obj))
+                    continue
+                ev = self._pending.pop(msg.get("id"), None)
+                if ev is not None:
+                    ev[1] = msg
+                    ev[0].set()
+        except (OSError, ValueError):
+            pass
+        finally:
+            self._closed = True
+            for ev in list(self._pending.values()):
+                ev[1] = {"ok": False, "error": {"type": "Error", "msg": "connection closed"}}
+                ev[0].set()
+            for w in self._watches.values():
+                w.stop()
+
+    def _call(self, verb: str, **kw):
+        if self._closed:
+            raise ConnectionError("netstore client closed")
+        with self._idlock:
+            self._next_id += 1
+            rid = self._next_id
+        ev = [threading.Event(), None]
+        self._pending[rid] = ev
+        req = json.dumps({"id": rid, "verb": verb, **kw}) + "\n"
+        with self._wlock:
+            self._sock.sendall(req.encode())
+        if not ev[0].wait(timeout=60.0):
+            self._pending.pop(rid, None)
+            raise TimeoutError(f"netstore call {verb} timed out")
+        msg = ev[1]
+        if msg["ok"]:
+            return msg.get("result")
+        err = msg["error"]
+        raise _EXC_BY_NAME.get(err["type"], RuntimeError)(err["msg"])
+
+    # -- MemStore interface -------------------------------------------------
+
+    def create(self, obj: dict) -> dict:
+        return self._call("create", obj=obj)
+
+    def get(self, kind: str, name: str, namespace: str = "") -> dict:
+        return self._call("get", kind=kind, name=name, namespace=namespace)
+
+    def list(self, kind: str, namespace: Optional[str] = None) -> List[dict]:
+        return self._call("list", kind=kind, namespace=namespace)
+
+    def update(self, obj: dict) -> dict:
+        return self._call("update", obj=obj)
+
+    def delete(self, kind: str, name: str, namespace: str = "", *, now: float = 0.0) -> None:
+        self._call("delete", kind=kind, name=name, namespace=namespace)
+
+    def watch(self, kind: Optional[str] = None, *, replay: bool = True):
+        w = _ClientWatch()
+        w.kind = kind
+        res = self._call("watch", kind=kind, replay=replay)
+        self._watches[res["watch_id"]] = w
+        return w
+
+    def update_with_retry(
+        self, kind: str, name: str, namespace: str,
+        mutate: Callable[[dict], Optional[dict]], attempts: int = 10,
+    ) -> Optional[dict]:
+        for _ in range(attempts):
+            try:
+                obj = self.get(kind, name, namespace)
+            except NotFound:
+                return None
+            new = mutate(obj)
+            if new is None:
+                return None
+            try:
+                return self.update(new)
+            except Conflict:
+                continue
+        raise Conflict(f"update_with_retry: {attempts} attempts exhausted for {kind}/{name}")
+
+    def close(self) -> None:
+        self._closed = True
+        try:
+            self._sock.close()
+        except OSError:
+            pass

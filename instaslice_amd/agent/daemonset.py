@@ -33,7 +33,14 @@ from instaslice_amd.api.types import (
 )
 from instaslice_amd.partition.profiles import catalog_from_amdsmi_profiles
 from instaslice_amd.runtime.engine import Engine, Key, Result, WatchSpec
-from instaslice_amd.smi.base import AmdSmi, PhysicalGpu, SmiBusy, SmiError, SmiNotSupported
+from instaslice_amd.smi.base import (
+    AmdSmi,
+    PhysicalGpu,
+    SmiBusy,
+    SmiError,
+    SmiNotSupported,
+    SmiPermission,
+)
 from instaslice_amd.store.memstore import AlreadyExists, MemStore, NotFound
 from instaslice_amd.utils import get_logger
 
@@ -63,6 +70,9 @@ class NodeAgent:
         self._gpus: Dict[str, PhysicalGpu] = {}
         # observability feed (metrics module scrapes this)
         self.reconfigure_events: List[dict] = []
+        # consecutive hard mode-set failures per GPU (3 strikes -> fail alloc)
+        self._mode_set_failures: Dict[str, int] = {}
+        self.prepare_failures = 0  # observability counter
         self.engine = Engine(
             name=f"agent-{node_name}",
             store=store,
@@ -266,6 +276,29 @@ class NodeAgent:
 
         self.store.update_with_retry("Node", self.node_name, "", mut)
 
+    def _needs_mode_change(self, alloc: dict) -> bool:
+        g = self._gpus.get(alloc["gpuUUID"])
+        return bool(g and g.compute_mode != alloc["computeMode"])
+
+    def _fail_allocation(self, pod_uuid: str, gpu_uuid: str, lock_mode: bool) -> None:
+        """Flip the allocation to `failed` (controller will re-place) and
+        optionally mark the GPU mode-locked in the CR."""
+
+        def mut(obj: dict) -> Optional[dict]:
+            spec = obj["spec"]
+            alloc = (spec.get("allocations") or {}).get(pod_uuid)
+            changed = False
+            if alloc and alloc["allocationStatus"] == AllocationStatus.CREATING:
+                alloc["allocationStatus"] = AllocationStatus.FAILED
+                changed = True
+            gd = (spec.get("gpus") or {}).get(gpu_uuid)
+            if lock_mode and gd and not gd.get("modeLocked"):
+                gd["modeLocked"] = True
+                changed = True
+            return obj if changed else None
+
+        self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, mut)
+
     # -- delete path ----------------------------------------------------------
 
     def _teardown_allocation(self, cr: dict, pod_uuid: str, alloc: dict) -> None:
@@ -288,12 +321,44 @@ class NodeAgent:
         for pod_uuid, alloc in sorted(allocations.items()):
             status = alloc["allocationStatus"]
             if status == AllocationStatus.CREATING:
+                needs_flip = self._needs_mode_change(alloc)
                 try:
                     prepared_entry = self._prepare_allocation(cr, alloc)
                 except SmiBusy as e:
                     self.log.warning("gpu busy, requeueing: %s", e)
                     requeue = REQUEUE_BUSY_S
                     continue
+                except (SmiNotSupported, SmiError) as e:
+                    # Hard device failure. The reference logs-and-continues
+                    # (instaslice_daemonset.go:173-189); we fail the
+                    # allocation so the controller re-places it — and after
+                    # repeated mode-set failures mark the GPU mode-locked so
+                    # the placer stops planning flips on it (VM guests).
+                    gpu = alloc["gpuUUID"]
+                    n = self._mode_set_failures.get(gpu, 0) + 1
+                    self._mode_set_failures[gpu] = n
+                    self.prepare_failures += 1
+                    # deterministic platform refusals lock the GPU's mode so
+                    # the placer stops planning flips on it (VM guests);
+                    # transient errors retry, then fail the allocation for
+                    # re-placement WITHOUT locking (self-healing)
+                    deterministic = isinstance(e, (SmiNotSupported, SmiPermission))
+                    if not deterministic and n < 3:
+                        self.log.warning(
+                            "prepare failed (attempt %d) for pod %s: %s",
+                            n, alloc["podName"], e,
+                        )
+                        requeue = REQUEUE_BUSY_S
+                        continue
+                    self.log.error(
+                        "prepare failed for pod %s on gpu %s (%s): failing allocation",
+                        alloc["podName"], gpu[:8], e,
+                    )
+                    self._mode_set_failures.pop(gpu, None)
+                    self._fail_allocation(pod_uuid, gpu,
+                                          lock_mode=deterministic and needs_flip)
+                    continue
+                self._mode_set_failures.pop(alloc["gpuUUID"], None)
                 if prepared_entry is None:
                     requeue = REQUEUE_BUSY_S
                     continue

@@ -24,14 +24,22 @@ from instaslice_amd import API_GROUP, API_VERSION
 
 
 class AllocationStatus:
-    """The 4-state protocol between controller and daemonset."""
+    """The status-string protocol between controller and daemonset
+    (SURVEY.md §1), extended with an explicit failure state the reference
+    lacks (it logs-and-ignores NVML errors, instaslice_daemonset.go:173-189):
+
+        creating -> created -> ungated -> deleted
+                 \\-> failed (daemonset could not realize; controller
+                              removes the allocation and re-places)
+    """
 
     CREATING = "creating"   # controller decided placement; daemonset must realize
     CREATED = "created"     # daemonset realized the partition
     UNGATED = "ungated"     # controller removed the pod's scheduling gate
     DELETED = "deleted"     # controller marked for teardown; daemonset must clean
+    FAILED = "failed"       # daemonset hit a hard device error; needs re-place
 
-    ALL = (CREATING, CREATED, UNGATED, DELETED)
+    ALL = (CREATING, CREATED, UNGATED, DELETED, FAILED)
 
 
 @dataclass

@@ -58,6 +58,9 @@ class GpuView:
     compute_mode: ComputeMode
     memory_mode: MemoryMode
     occupied: Set[int]  # partition ordinals referenced by allocations/prepared
+    # the daemonset hit a hard error flipping this GPU's mode (e.g. VM guest
+    # forbids it): only same-mode placements are possible from now on
+    mode_locked: bool = False
 
 
 def build_gpu_views(node_name: str, spec: dict) -> List[GpuView]:
@@ -97,6 +100,7 @@ def build_gpu_views(node_name: str, spec: dict) -> List[GpuView]:
                 compute_mode=ComputeMode(compute),
                 memory_mode=MemoryMode(memory),
                 occupied=occupied.get(uuid, set()),
+                mode_locked=bool(gd.get("modeLocked")),
             )
         )
     views.sort(key=lambda v: v.index)
@@ -142,8 +146,8 @@ def _free_ordinal(view: GpuView, n_partitions: int) -> Optional[int]:
 def _place_on(view: GpuView, profile: PartitionProfile, needs_change: bool) -> Optional[Placement]:
     n = profile.partitions_per_gpu
     if needs_change:
-        if view.occupied:
-            return None  # mode change needs an idle GPU
+        if view.occupied or view.mode_locked:
+            return None  # mode change needs an idle, flippable GPU
         mem, _ = _target_memory_mode(profile, view.memory_mode)
         return Placement(view.node, view.uuid, 0, profile.compute.value, mem.value, True)
     k = _free_ordinal(view, n)

@@ -81,7 +81,10 @@ class PodController:
     def _instaslice_to_pods(self, event_type: str, obj: dict) -> List[Key]:
         keys: List[Key] = []
         for alloc in (obj.get("spec", {}).get("allocations") or {}).values():
-            if alloc.get("allocationStatus") == AllocationStatus.CREATED:
+            if alloc.get("allocationStatus") in (
+                AllocationStatus.CREATED,
+                AllocationStatus.FAILED,
+            ):
                 keys.append(("Pod", alloc["namespace"], alloc["podName"]))
         return keys
 
@@ -256,6 +259,26 @@ class PodController:
             if t0 is not None:
                 self.alloc_latency_s.append(time.monotonic() - t0)
             return Result()
+        if status == AllocationStatus.FAILED:
+            # daemonset could not realize it: drop the allocation and
+            # re-place on the next pass (the CR now carries modeLocked hints
+            # so the placer avoids the failing transition)
+            def drop(cr_obj: dict) -> Optional[dict]:
+                allocs = cr_obj.get("spec", {}).get("allocations") or {}
+                a = allocs.get(uid)
+                if not a or a["allocationStatus"] != AllocationStatus.FAILED:
+                    return None
+                del allocs[uid]
+                return cr_obj
+
+            self.store.update_with_retry(
+                "Instaslice", cr["metadata"]["name"], INSTASLICE_NS, drop
+            )
+            self.log.warning(
+                "allocation for pod %s failed on %s; re-placing",
+                name, alloc["gpuUUID"][:8],
+            )
+            return Result(requeue_after=0.01)
         # creating / ungated / deleted: nothing to do here
         return Result()
 

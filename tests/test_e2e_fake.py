@@ -163,9 +163,9 @@ def test_smi_failure_fails_allocation_loudly():
         smi.fault_hook = hook
         c.submit_pod("p1", "cpx-1x36")
         with pytest.raises(TimeoutError):
-            c.wait_pod_scheduled("p1", timeout=0.5)
-        assert c.agents["node-0"].engine.error_count > 0
-        # heal the fault: the engine retries and the pod schedules
+            c.wait_pod_scheduled("p1", timeout=1.0)
+        assert c.agents["node-0"].prepare_failures > 0
+        # heal the fault: retry/re-place machinery schedules the pod
         smi.fault_hook = None
         c.wait_pod_scheduled("p1", timeout=10.0)
     finally:
@@ -223,3 +223,36 @@ def test_agent_restart_adopts_state(cluster):
     ords = sorted(p["ordinal"] for p in cluster.prepared("node-0").values())
     assert ords == [0, 1]
     new.stop()
+
+
+def test_mode_locked_gpu_replacement():
+    """A GPU that deterministically refuses mode flips (VM guest) gets marked
+    modeLocked; the controller re-places the pod on another GPU."""
+    from instaslice_amd.smi import SmiNotSupported
+
+    c = Cluster(teardown_grace_s=0.0)
+    smi = FakeAmdSmi(num_gpus=2, node_name="node-0")
+    locked_uuid = {}
+
+    def hook(verb, gpu):
+        if verb == "set_compute_partition" and gpu == locked_uuid.get("u"):
+            raise SmiNotSupported("platform forbids partitioning on gpu0")
+
+    c.add_node("node-0", smi)
+    c.start()
+    try:
+        gpus = smi.list_gpus()
+        locked_uuid["u"] = gpus[0].uuid
+        smi.fault_hook = hook
+        c.submit_pod("p1", "cpx-1x36")
+        c.wait_pod_scheduled("p1", timeout=15.0)
+        # landed on gpu1, and gpu0 is now marked mode-locked in the CR
+        (alloc,) = c.allocations("node-0").values()
+        assert alloc["gpuUUID"] == gpus[1].uuid
+        cr = c.store.get("Instaslice", "node-0", "instaslice-system")
+        assert cr["spec"]["gpus"][gpus[0].uuid].get("modeLocked") is True
+        # next pod goes straight to gpu1 without retrying gpu0
+        c.submit_pod("p2", "cpx-1x36")
+        c.wait_pod_scheduled("p2", timeout=15.0)
+    finally:
+        c.stop()

@@ -230,13 +230,18 @@ class NodeAgent:
         # the workload contract: a pod-named ConfigMap delivering the visible
         # partition (reference: createConfigMap with NVIDIA_/CUDA_VISIBLE_DEVICES,
         # instaslice_daemonset.go:796-818 -> ROCR_/HIP_ here)
+        # ROCR accepts device ordinals (or rocminfo-style GPU-<id> strings),
+        # NOT amdsmi UUIDs — verified on MI355X: a UUID value yields "no
+        # ROCm-capable device is detected". The partition's node-wide HIP
+        # ordinal is the selector; the amdsmi UUID rides along for operators.
         cm = {
             "apiVersion": "v1",
             "kind": "ConfigMap",
             "metadata": {"name": alloc["podName"], "namespace": alloc["namespace"]},
             "data": {
-                "ROCR_VISIBLE_DEVICES": part.uuid,
+                "ROCR_VISIBLE_DEVICES": str(part.device_index),
                 "HIP_VISIBLE_DEVICES": str(part.device_index),
+                "INSTASLICE_PARTITION_UUID": part.uuid,
                 "INSTASLICE_PARTITION_ORDINAL": str(ordinal),
                 "INSTASLICE_PARTITION_GPU": gpu_uuid,
             },

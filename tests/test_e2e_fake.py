@@ -27,7 +27,8 @@ def test_happy_path_single_pod(cluster):
     pod = cluster.wait_pod_scheduled("p1")
     assert not pod["spec"]["schedulingGates"]
     env = cluster.pod_env("p1")
-    assert "ROCR_VISIBLE_DEVICES" in env and env["ROCR_VISIBLE_DEVICES"]
+    assert env["ROCR_VISIBLE_DEVICES"] == "0"  # HIP ordinal (ROCR selector)
+    assert env["INSTASLICE_PARTITION_UUID"]
     assert env["INSTASLICE_PARTITION_ORDINAL"] == "0"
     allocs = cluster.allocations("node-0")
     (alloc,) = allocs.values()
@@ -70,9 +71,10 @@ def test_eight_cpx_pods_fill_one_gpu(cluster):
     ordinals = sorted(p["ordinal"] for p in prepared.values())
     assert ordinals == list(range(8))
     assert len(cluster.agents["node-0"].reconfigure_events) == 1
-    # distinct visible devices per pod
-    uuids = {cluster.pod_env(f"p{i}")["ROCR_VISIBLE_DEVICES"] for i in range(8)}
-    assert len(uuids) == 8
+    # distinct visible devices per pod (both ordinal and partition uuid)
+    ords = {cluster.pod_env(f"p{i}")["ROCR_VISIBLE_DEVICES"] for i in range(8)}
+    uuids = {cluster.pod_env(f"p{i}")["INSTASLICE_PARTITION_UUID"] for i in range(8)}
+    assert len(ords) == 8 and len(uuids) == 8
 
 
 def test_mixed_profiles_across_gpus(cluster):

@@ -206,6 +206,11 @@ class NodeAgent:
             except SmiError:
                 event["metrics_after"] = {}
             self.reconfigure_events.append(event)
+            from instaslice_amd.metrics import get_metrics
+
+            get_metrics().reconfigure(
+                self.node_name, f"{compute}/{memory}", event["set_wall_ms"] / 1000.0
+            )
             g = self._gpus[gpu_uuid]
             self.log.info(
                 "gpu %s reconfigured %s -> %s/%s in %.1f ms (%d partitions)",
@@ -303,6 +308,9 @@ class NodeAgent:
             return obj if changed else None
 
         self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, mut)
+        from instaslice_amd.metrics import get_metrics
+
+        get_metrics().allocation("failed")
 
     # -- delete path ----------------------------------------------------------
 
@@ -388,6 +396,9 @@ class NodeAgent:
                     return obj
 
                 self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, commit)
+                from instaslice_amd.metrics import get_metrics
+
+                get_metrics().allocation("created")
                 self.log.debug(
                     "prepared partition %s for pod %s", part_uuid[:8], alloc["podName"]
                 )
@@ -410,6 +421,9 @@ class NodeAgent:
                     return obj
 
                 self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, cleanup)
+                from instaslice_amd.metrics import get_metrics
+
+                get_metrics().allocation("deleted")
                 if self.reset_mode_on_empty:
                     self._maybe_reset_gpu(alloc["gpuUUID"])
         return Result(requeue_after=requeue)

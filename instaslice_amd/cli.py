@@ -1,0 +1,245 @@
+"""Operational entry points — the analog of the reference's two binaries
+(cmd/controller/main.go, cmd/daemonset/main.go) plus operator conveniences.
+
+    python -m instaslice_amd store      [--port 7080]
+    python -m instaslice_amd controller --store HOST:PORT [--policy packed-fit]
+                                        [--metrics-port 8080] [--grace 30]
+                                        [--leader-elect]
+    python -m instaslice_amd daemonset  --store HOST:PORT [--node-name NAME]
+                                        [--fake N] [--metrics-port 8084]
+                                        [--reset-on-empty]
+    python -m instaslice_amd submit     --store HOST:PORT --name P --profile PR
+                                        [--node NODE] [--wait]
+    python -m instaslice_amd delete     --store HOST:PORT --name P
+    python -m instaslice_amd status     --store HOST:PORT
+    python -m instaslice_amd payload    [info|vecadd|membw|busy|census] ...
+
+NODE_NAME env is honored for the daemonset (downward-API parity with
+config/manager/manager.yaml's fieldRef env).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import signal
+import socket
+import sys
+import time
+
+from instaslice_amd.utils import get_logger
+
+log = get_logger("cli")
+
+
+def _connect(store_arg: str):
+    from instaslice_amd.store.netstore import NetStoreClient
+
+    host, _, port = store_arg.rpartition(":")
+    return NetStoreClient(host or "127.0.0.1", int(port))
+
+
+def _wait_forever():
+    stop = []
+    signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
+    signal.signal(signal.SIGINT, lambda *a: stop.append(1))
+    while not stop:
+        time.sleep(0.5)
+
+
+def cmd_store(args) -> int:
+    from instaslice_amd.store.netstore import StoreServer
+
+    server = StoreServer(port=args.port).start()
+    log.info("store serving on %s:%d", server.host, server.port)
+    _wait_forever()
+    server.stop()
+    return 0
+
+
+def cmd_controller(args) -> int:
+    from instaslice_amd.controller.reconciler import PodController
+    from instaslice_amd.metrics import get_metrics, serve_http
+
+    store = _connect(args.store)
+    lease = None
+    if args.leader_elect:
+        from instaslice_amd.runtime.lease import LeaderLease
+
+        lease = LeaderLease(store, "instaslice-controller").start()
+        log.info("waiting for leadership...")
+        lease.wait_leader()
+    controller = PodController(store, policy=args.policy,
+                               teardown_grace_s=args.grace)
+    controller.start()
+    srv = None
+    if args.metrics_port:
+        srv = serve_http(get_metrics(), args.metrics_port)
+        log.info("metrics on :%d", args.metrics_port)
+    log.info("controller running (policy=%s grace=%.0fs)", args.policy, args.grace)
+    _wait_forever()
+    controller.stop()
+    if srv:
+        srv.shutdown()
+    if lease:
+        lease.stop()
+    return 0
+
+
+def cmd_daemonset(args) -> int:
+    from instaslice_amd.agent.daemonset import NodeAgent
+    from instaslice_amd.metrics import get_metrics, serve_http
+
+    node = args.node_name or os.environ.get("NODE_NAME") or socket.gethostname()
+    store = _connect(args.store)
+    if args.fake:
+        from instaslice_amd.smi.fake import FakeAmdSmi
+
+        smi = FakeAmdSmi(num_gpus=args.fake, node_name=node)
+    else:
+        from instaslice_amd.smi.native import NativeAmdSmi
+
+        smi = NativeAmdSmi()
+    agent = NodeAgent(store, smi, node, reset_mode_on_empty=args.reset_on_empty)
+    agent.start()
+    srv = None
+    if args.metrics_port:
+        srv = serve_http(get_metrics(), args.metrics_port)
+        log.info("metrics on :%d", args.metrics_port)
+    log.info("daemonset running on node %s (%s)", node,
+             "fake" if args.fake else "amdsmi")
+    _wait_forever()
+    agent.stop()
+    if srv:
+        srv.shutdown()
+    return 0
+
+
+def cmd_submit(args) -> int:
+    from instaslice_amd.api.types import new_pod
+
+    store = _connect(args.store)
+    sel = {"kubernetes.io/hostname": args.node} if args.node else None
+    pod = new_pod(args.name, namespace=args.namespace, profile=args.profile,
+                  node_selector=sel)
+    store.create(pod)
+    print(f"pod {args.namespace}/{args.name} submitted (gated)")
+    if args.wait:
+        deadline = time.monotonic() + args.timeout
+        while time.monotonic() < deadline:
+            p = store.get("Pod", args.name, args.namespace)
+            if not p["spec"].get("schedulingGates"):
+                cm = store.get("ConfigMap", args.name, args.namespace)
+                print(f"scheduled; env: {json.dumps(cm['data'])}")
+                return 0
+            time.sleep(0.05)
+        print("timed out waiting for scheduling", file=sys.stderr)
+        return 1
+    return 0
+
+
+def cmd_delete(args) -> int:
+    store = _connect(args.store)
+    store.delete("Pod", args.name, args.namespace)
+    print(f"pod {args.namespace}/{args.name} deletion requested")
+    return 0
+
+
+def cmd_status(args) -> int:
+    store = _connect(args.store)
+    out = {"nodes": [], "pods": []}
+    for cr in store.list("Instaslice"):
+        spec = cr.get("spec", {})
+        out["nodes"].append({
+            "node": cr["metadata"]["name"],
+            "gpus": {
+                u[:8]: {
+                    "mode": f"{g.get('computeMode')}/{g.get('memoryMode')}",
+                    "used": g.get("usedOrdinals", []),
+                    "modeLocked": g.get("modeLocked", False),
+                }
+                for u, g in (spec.get("gpus") or {}).items()
+            },
+            "allocations": {
+                a["podName"]: f"{a['profile']}@{a['gpuUUID'][:8]}#{a['ordinal']} "
+                              f"({a['allocationStatus']})"
+                for a in (spec.get("allocations") or {}).values()
+            },
+            "prepared": len(spec.get("prepared") or {}),
+        })
+    for pod in store.list("Pod"):
+        out["pods"].append({
+            "name": pod["metadata"]["name"],
+            "gated": bool(pod["spec"].get("schedulingGates")),
+            "deleting": bool(pod["metadata"].get("deletionTimestamp")),
+        })
+    print(json.dumps(out, indent=2))
+    return 0
+
+
+def cmd_payload(args) -> int:
+    import subprocess
+
+    bin_path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "bin", "instaslice-payload")
+    return subprocess.call([bin_path] + args.payload_args)
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser(prog="instaslice-amd")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    p = sub.add_parser("store", help="run the state store server")
+    p.add_argument("--port", type=int, default=7080)
+    p.set_defaults(fn=cmd_store)
+
+    p = sub.add_parser("controller", help="run the cluster controller")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--policy", default="packed-fit")
+    p.add_argument("--grace", type=float, default=30.0,
+                   help="teardown grace seconds (reference: 30)")
+    p.add_argument("--metrics-port", type=int, default=8080)
+    p.add_argument("--leader-elect", action="store_true")
+    p.set_defaults(fn=cmd_controller)
+
+    p = sub.add_parser("daemonset", help="run the per-node agent")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--node-name", default=None)
+    p.add_argument("--fake", type=int, default=0,
+                   help="use FakeAmdSmi with N GPUs instead of libamd_smi")
+    p.add_argument("--metrics-port", type=int, default=8084)
+    p.add_argument("--reset-on-empty", action="store_true",
+                   help="return drained GPUs to SPX/NPS1 (reference parity)")
+    p.set_defaults(fn=cmd_daemonset)
+
+    p = sub.add_parser("submit", help="submit a gated pod requesting a partition")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--name", required=True)
+    p.add_argument("--namespace", default="default")
+    p.add_argument("--profile", required=True, help="e.g. cpx-1x36")
+    p.add_argument("--node", default=None)
+    p.add_argument("--wait", action="store_true")
+    p.add_argument("--timeout", type=float, default=60.0)
+    p.set_defaults(fn=cmd_submit)
+
+    p = sub.add_parser("delete", help="delete a pod (starts teardown)")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--name", required=True)
+    p.add_argument("--namespace", default="default")
+    p.set_defaults(fn=cmd_delete)
+
+    p = sub.add_parser("status", help="cluster state summary")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.set_defaults(fn=cmd_status)
+
+    p = sub.add_parser("payload", help="run the HIP payload binary")
+    p.add_argument("payload_args", nargs="*", default=["info"])
+    p.set_defaults(fn=cmd_payload)
+
+    args = ap.parse_args(argv)
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

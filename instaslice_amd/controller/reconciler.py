@@ -257,7 +257,11 @@ class PodController:
             )
             t0 = self._pending_since.pop(uid, None)
             if t0 is not None:
-                self.alloc_latency_s.append(time.monotonic() - t0)
+                dt = time.monotonic() - t0
+                self.alloc_latency_s.append(dt)
+                from instaslice_amd.metrics import get_metrics
+
+                get_metrics().allocation_latency(dt)
             return Result()
         if status == AllocationStatus.FAILED:
             # daemonset could not realize it: drop the allocation and

@@ -121,19 +121,27 @@ class Engine:
                 self.enqueue(k)
 
     def _work(self) -> None:
+        from instaslice_amd.metrics import get_metrics
+
+        metrics = get_metrics()
         while True:
             key = self._next_key()
             if key is None:
                 return
+            t0 = time.monotonic()
             try:
                 self.reconcile_count += 1
                 res = self.reconcile(key)
             except Exception:
                 self.error_count += 1
+                metrics.reconcile(self.name, "error", time.monotonic() - t0)
                 self.log.error("reconcile %s failed:\n%s", key, traceback.format_exc())
                 self.enqueue_after(key, self.error_backoff_s)
                 continue
-            if res and res.requeue_after is not None:
+            requeued = res and res.requeue_after is not None
+            metrics.reconcile(self.name, "requeue" if requeued else "ok",
+                              time.monotonic() - t0)
+            if requeued:
                 self.enqueue_after(key, res.requeue_after)
 
     # -- lifecycle ----------------------------------------------------------

@@ -96,10 +96,14 @@ class _Handler(socketserver.StreamRequestHandler):
                         watches.append(w)
                         next_watch_id += 1
                         wid = next_watch_id
+                        # respond BEFORE pumping so the id precedes any event
+                        # on the wire (the client still buffers orphans in
+                        # case its caller hasn't registered the id yet)
+                        send({"id": rid, "ok": True, "result": {"watch_id": wid}})
                         threading.Thread(
                             target=pump_watch, args=(wid, w), daemon=True
                         ).start()
-                        res = {"watch_id": wid}
+                        continue
                     elif verb == "ping":
                         res = "pong"
                     else:
@@ -177,6 +181,11 @@ class NetStoreClient:
         self._wlock = threading.Lock()
         self._pending: dict = {}
         self._watches: dict = {}
+        # events that arrived before watch() registered their id (the reader
+        # thread outruns the caller between response and registration);
+        # _watch_reg_lock makes register-vs-deliver atomic
+        self._orphan_events: dict = {}
+        self._watch_reg_lock = threading.Lock()
         self._next_id = 0
         self._idlock = threading.Lock()
         self._closed = False
@@ -190,9 +199,13 @@ class NetStoreClient:
             for line in self._rfile:
                 msg = json.loads(line)
                 if "watch_id" in msg and "event" in msg:
-                    w = self._watches.get(msg["watch_id"])
-                    if w and not w._stopped:
-                        et, obj = msg["event"]
+                    wid = msg["watch_id"]
+                    et, obj = msg["event"]
+                    with self._watch_reg_lock:
+                        w = self._watches.get(wid)
+                        if w is None:
+                            self._orphan_events.setdefault(wid, []).append((et, obj))
+                    if w is not None and not w._stopped:
                         w._q.put((et, obj))
                     continue
                 ev = self._pending.pop(msg.get("id"), None)
@@ -250,7 +263,13 @@ class NetStoreClient:
         w = _ClientWatch()
         w.kind = kind
         res = self._call("watch", kind=kind, replay=replay)
-        self._watches[res["watch_id"]] = w
+        wid = res["watch_id"]
+        with self._watch_reg_lock:
+            self._watches[wid] = w
+            orphans = self._orphan_events.pop(wid, [])
+        # deliver events that raced ahead of this registration, in order
+        for ev in orphans:
+            w._q.put(ev)
         return w
 
     def update_with_retry(

@@ -1,0 +1,98 @@
+"""CLI + leader-election tests: the full operational topology as processes
+would run it (store server, controller, fake daemonset, submit/status)."""
+
+import json
+import subprocess
+import sys
+import time
+
+import pytest
+
+from instaslice_amd.runtime.lease import LeaderLease
+from instaslice_amd.store.memstore import MemStore
+from instaslice_amd.store.netstore import NetStoreClient, StoreServer
+
+
+def test_leader_election_single_winner():
+    store = MemStore()
+    a = LeaderLease(store, "ctl", identity="a", ttl_s=1.0, renew_every_s=0.1).start()
+    b = LeaderLease(store, "ctl", identity="b", ttl_s=1.0, renew_every_s=0.1).start()
+    try:
+        assert a.wait_leader(2.0) or b.wait_leader(2.0)
+        time.sleep(0.3)
+        assert a.is_leader.is_set() != b.is_leader.is_set()  # exactly one
+        winner, loser = (a, b) if a.is_leader.is_set() else (b, a)
+        winner.stop()  # releases
+        assert loser.wait_leader(3.0), "successor never acquired"
+    finally:
+        a.stop()
+        b.stop()
+
+
+def test_cli_end_to_end_processes():
+    """store + controller + fake daemonset as real subprocesses, driven by
+    submit/status/delete — the deployment shape of the k8s manifests."""
+    server = StoreServer().start()
+    addr = f"127.0.0.1:{server.port}"
+    procs = []
+
+    def spawn(*cmd):
+        p = subprocess.Popen([sys.executable, "-m", "instaslice_amd", *cmd],
+                             stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                             text=True)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("controller", "--store", addr, "--grace", "0", "--metrics-port", "0")
+        spawn("daemonset", "--store", addr, "--node-name", "cli-node",
+              "--fake", "2", "--metrics-port", "0")
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "submit", "--store", addr,
+             "--name", "clipod", "--profile", "qpx-2x72", "--wait",
+             "--timeout", "30"],
+            capture_output=True, text=True, timeout=60)
+        if out.returncode != 0:
+            import json as _json
+            diag = [out.stdout, out.stderr,
+                    "CRs: " + _json.dumps(server.store.list("Instaslice"))[:1500],
+                    "Pods: " + _json.dumps(server.store.list("Pod"))[:800]]
+            for p_ in procs:
+                p_.terminate()
+                try:
+                    o, _ = p_.communicate(timeout=5)
+                except subprocess.TimeoutExpired:
+                    p_.kill()
+                    o, _ = p_.communicate()
+                diag.append(o[-3000:] if o else "<no output>")
+            pytest.fail("submit failed:\n" + "\n=====\n".join(diag))
+        assert "ROCR_VISIBLE_DEVICES" in out.stdout
+
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "status", "--store", addr],
+            capture_output=True, text=True, timeout=30)
+        st = json.loads(out.stdout)
+        assert st["nodes"][0]["node"] == "cli-node"
+        assert "clipod" in st["nodes"][0]["allocations"]
+
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "delete", "--store", addr,
+             "--name", "clipod"], capture_output=True, text=True, timeout=30)
+        assert out.returncode == 0
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            crs = server.store.list("Instaslice")
+            if crs and not crs[0]["spec"].get("allocations"):
+                break
+            time.sleep(0.05)
+        else:
+            pytest.fail("teardown did not drain")
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
+        server.stop()

@@ -42,3 +42,7 @@ GATE_NAME = "org.instaslice/accelarator"
 FINALIZER_NAME = "org.instaslice/accelarator"
 RESOURCE_PREFIX = "amd.com/"  # e.g. limits: {"amd.com/cpx-1x36": 1}
 POD_RESOURCE_PREFIX = "org.instaslice/"  # per-pod extended resource pinning
+# set on a gated pod when no node can currently fit its profile (the
+# reference requeues silently, instaslice_controller.go:231; surfacing the
+# condition lets schedulers/users react instead of waiting blind)
+UNSCHEDULABLE_ANNOTATION = "org.instaslice/unschedulable"

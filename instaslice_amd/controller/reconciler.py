@@ -64,6 +64,9 @@ class PodController:
         # latency observability: pod_uid -> submit time, and measured p50 feed
         self.alloc_latency_s: List[float] = []
         self._pending_since: dict = {}
+        # pods currently marked unschedulable: re-reconciled on ANY node-state
+        # change so freed capacity is picked up event-driven, not by polling
+        self._unschedulable_keys: set = set()
         self.engine = Engine(
             name="controller",
             store=store,
@@ -86,6 +89,8 @@ class PodController:
                 AllocationStatus.FAILED,
             ):
                 keys.append(("Pod", alloc["namespace"], alloc["podName"]))
+        # freed/changed capacity: give waiting pods another placement pass
+        keys.extend(("Pod", ns, name) for (ns, name) in list(self._unschedulable_keys))
         return keys
 
     # -- helpers ------------------------------------------------------------
@@ -172,8 +177,36 @@ class PodController:
                     md["name"], profile_name, node_name,
                     placement.gpu_uuid[:8], placement.ordinal,
                 )
+                self._mark_unschedulable(md, profile_name, unschedulable=False)
                 return Result()  # wait for agent's "created" event
+        self._mark_unschedulable(md, profile_name, unschedulable=True)
         return Result(requeue_after=self.requeue_no_fit_s)
+
+    def _mark_unschedulable(self, md: dict, profile_name: str,
+                            unschedulable: bool) -> None:
+        from instaslice_amd import UNSCHEDULABLE_ANNOTATION
+
+        key = (md.get("namespace", "default"), md["name"])
+        if unschedulable:
+            self._unschedulable_keys.add(key)
+        else:
+            self._unschedulable_keys.discard(key)
+
+        def mut(p: dict) -> Optional[dict]:
+            ann = p["metadata"].setdefault("annotations", {})
+            if unschedulable:
+                if ann.get(UNSCHEDULABLE_ANNOTATION) == profile_name:
+                    return None
+                ann[UNSCHEDULABLE_ANNOTATION] = profile_name
+            else:
+                if UNSCHEDULABLE_ANNOTATION not in ann:
+                    return None
+                del ann[UNSCHEDULABLE_ANNOTATION]
+            return p
+
+        self.store.update_with_retry(
+            "Pod", md["name"], md.get("namespace", "default"), mut
+        )
 
     def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str) -> None:
         def mut(cr: dict) -> Optional[dict]:
@@ -192,12 +225,14 @@ class PodController:
         try:
             pod = self.store.get("Pod", name, namespace)
         except NotFound:
+            self._unschedulable_keys.discard((namespace, name))
             return Result()
         md = pod["metadata"]
         uid = md["uid"]
 
         # teardown path (reference: instaslice_controller.go:99-142)
         if md.get("deletionTimestamp"):
+            self._unschedulable_keys.discard((namespace, name))
             gated = pod_is_gated(pod)
             if not gated:
                 elapsed = time.time() - float(md["deletionTimestamp"])

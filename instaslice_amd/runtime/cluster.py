@@ -83,6 +83,27 @@ class Cluster:
             time.sleep(0.002)
         raise TimeoutError(f"pod {namespace}/{name} not scheduled in {timeout}s")
 
+    def wait_pod_outcome(self, name: str, namespace: str = "default",
+                         timeout: float = 10.0) -> str:
+        """Block until the pod is either scheduled ("scheduled") or marked
+        unschedulable by the controller ("unschedulable")."""
+        from instaslice_amd import UNSCHEDULABLE_ANNOTATION
+
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            try:
+                pod = self.store.get("Pod", name, namespace)
+            except NotFound:
+                time.sleep(0.002)
+                continue
+            if not pod["spec"].get("schedulingGates"):
+                return "scheduled"
+            ann = pod["metadata"].get("annotations") or {}
+            if UNSCHEDULABLE_ANNOTATION in ann:
+                return "unschedulable"
+            time.sleep(0.002)
+        raise TimeoutError(f"pod {namespace}/{name}: no outcome in {timeout}s")
+
     def delete_pod(self, name: str, namespace: str = "default") -> None:
         self.store.delete("Pod", name, namespace)
 

@@ -1,0 +1,82 @@
+"""Packaging-tier tests: manifests parse, the CRD stays generated, and the
+pod contract in samples matches what the code implements."""
+
+import subprocess
+import sys
+from pathlib import Path
+
+import yaml
+
+ROOT = Path(__file__).resolve().parent.parent
+
+
+def _load_all(path):
+    return [d for d in yaml.safe_load_all(path.read_text()) if d]
+
+
+def test_crd_matches_generator():
+    """config/crd must equal api/crd.py output (controller-gen analog)."""
+    from instaslice_amd.api.crd import crd_manifest
+
+    path = ROOT / "config/crd/bases/inference.codeflare.dev_instaslices.yaml"
+    (checked_in,) = _load_all(path)
+    assert checked_in == crd_manifest(), (
+        "CRD manifest drifted; regenerate with `make crd`"
+    )
+
+
+def test_crd_schema_accepts_real_cr():
+    """The schema's required fields are satisfiable by what the agent writes."""
+    from instaslice_amd.api.crd import ALLOCATION_SCHEMA, PREPARED_SCHEMA
+    from instaslice_amd.api.types import AllocationDetails, PreparedDetails
+
+    alloc = AllocationDetails(
+        profile="cpx-1x36", gpu_uuid="u", ordinal=0, start=0, size=1,
+        pod_uuid="p", pod_name="n", namespace="default", nodename="node",
+    ).to_dict()
+    for req in ALLOCATION_SCHEMA["required"]:
+        assert req in alloc
+    assert alloc["allocationStatus"] in ALLOCATION_SCHEMA["properties"][
+        "allocationStatus"]["enum"]
+    prep = PreparedDetails(
+        parent_gpu_uuid="u", ordinal=0, compute_mode="CPX", memory_mode="NPS1",
+        xcds=1, memory_gb=36, pod_uuid="p",
+    ).to_dict()
+    for req in PREPARED_SCHEMA["required"]:
+        assert req in prep
+
+
+def test_all_manifests_parse():
+    for p in (ROOT / "config").rglob("*.yaml"):
+        _load_all(p)
+    for p in (ROOT / "samples").rglob("*.yaml"):
+        _load_all(p)
+    for p in (ROOT / "deploy").rglob("*.yaml"):
+        _load_all(p)
+
+
+def test_sample_pod_contract():
+    """samples/test-pod.yaml follows the contract the controller expects."""
+    from instaslice_amd import FINALIZER_NAME, GATE_NAME
+    from instaslice_amd.partition.profiles import extract_profile_from_limits
+
+    (pod,) = _load_all(ROOT / "samples/test-pod.yaml")
+    assert any(g["name"] == GATE_NAME for g in pod["spec"]["schedulingGates"])
+    assert FINALIZER_NAME in pod["metadata"]["finalizers"]
+    limits = pod["spec"]["containers"][0]["resources"]["limits"]
+    profile = extract_profile_from_limits(limits)
+    assert profile == "cpx-1x36"
+    # pod-named extended resource + pod-named ConfigMap
+    name = pod["metadata"]["name"]
+    assert limits.get(f"org.instaslice/{name}") == 1
+    assert pod["spec"]["containers"][0]["envFrom"][0]["configMapRef"]["name"] == name
+
+
+def test_rbac_covers_all_kinds_the_code_touches():
+    (role,) = _load_all(ROOT / "config/rbac/role.yaml")
+    covered = set()
+    for rule in role["rules"]:
+        for r in rule["resources"]:
+            covered.add(r)
+    for needed in ("pods", "configmaps", "nodes", "instaslices", "leases"):
+        assert needed in covered, f"RBAC misses {needed}"

@@ -60,6 +60,11 @@ class Engine:
 
         self._queue: List[Key] = []
         self._queued: set = set()
+        # keys currently being reconciled (multi-worker safety: a key is
+        # never processed concurrently; events arriving mid-reconcile mark it
+        # dirty and it re-runs right after — controller-runtime semantics)
+        self._in_flight: set = set()
+        self._dirty: set = set()
         self._timers: List[Tuple[float, int, Key]] = []  # heap by deadline
         self._timer_seq = 0
         self._cv = threading.Condition()
@@ -74,7 +79,9 @@ class Engine:
 
     def enqueue(self, key: Key) -> None:
         with self._cv:
-            if key not in self._queued:
+            if key in self._in_flight:
+                self._dirty.add(key)
+            elif key not in self._queued:
                 self._queue.append(key)
                 self._queued.add(key)
                 self._cv.notify_all()
@@ -97,6 +104,7 @@ class Engine:
                 if self._queue:
                     k = self._queue.pop(0)
                     self._queued.discard(k)
+                    self._in_flight.add(k)
                     return k
                 timeout = None
                 if self._timers:
@@ -136,13 +144,27 @@ class Engine:
                 self.error_count += 1
                 metrics.reconcile(self.name, "error", time.monotonic() - t0)
                 self.log.error("reconcile %s failed:\n%s", key, traceback.format_exc())
+                self._finish(key, dirty_requeue=True)
                 self.enqueue_after(key, self.error_backoff_s)
                 continue
             requeued = res and res.requeue_after is not None
             metrics.reconcile(self.name, "requeue" if requeued else "ok",
                               time.monotonic() - t0)
+            self._finish(key)
             if requeued:
                 self.enqueue_after(key, res.requeue_after)
+
+    def _finish(self, key: Key, dirty_requeue: bool = False) -> None:
+        with self._cv:
+            self._in_flight.discard(key)
+            if key in self._dirty and not dirty_requeue:
+                self._dirty.discard(key)
+                if key not in self._queued:
+                    self._queue.append(key)
+                    self._queued.add(key)
+                    self._cv.notify_all()
+            else:
+                self._dirty.discard(key)
 
     # -- lifecycle ----------------------------------------------------------
 
@@ -178,7 +200,8 @@ class Engine:
         quiet_since = None
         while time.monotonic() < deadline:
             with self._cv:
-                empty = not self._queue and not self._queued
+                empty = (not self._queue and not self._queued
+                         and not self._in_flight)
             if empty:
                 if quiet_since is None:
                     quiet_since = time.monotonic()

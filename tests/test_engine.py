@@ -1,0 +1,106 @@
+"""Reconcile-engine semantics: dedup, requeue-after, multi-worker key safety."""
+
+import threading
+import time
+
+from instaslice_amd.runtime.engine import Engine, Result, WatchSpec
+from instaslice_amd.store.memstore import MemStore
+
+
+def _obj(name):
+    return {"apiVersion": "v1", "kind": "Thing",
+            "metadata": {"name": name, "namespace": ""}}
+
+
+def test_requeue_after_fires():
+    store = MemStore()
+    seen = []
+
+    def rec(key):
+        seen.append(time.monotonic())
+        if len(seen) < 3:
+            return Result(requeue_after=0.05)
+        return Result()
+
+    eng = Engine("t", store, rec, [WatchSpec(kind="Thing")]).start()
+    try:
+        store.create(_obj("a"))
+        deadline = time.monotonic() + 5
+        while len(seen) < 3 and time.monotonic() < deadline:
+            time.sleep(0.01)
+        assert len(seen) >= 3
+        assert seen[1] - seen[0] >= 0.04
+    finally:
+        eng.stop()
+
+
+def test_no_concurrent_same_key_with_multiple_workers():
+    store = MemStore()
+    active = {}
+    overlaps = []
+    lock = threading.Lock()
+    done = threading.Event()
+    calls = [0]
+
+    def rec(key):
+        with lock:
+            if active.get(key):
+                overlaps.append(key)
+            active[key] = True
+        time.sleep(0.01)  # widen the race window
+        with lock:
+            active[key] = False
+            calls[0] += 1
+            if calls[0] >= 30:
+                done.set()
+        return Result()
+
+    eng = Engine("t", store, rec, [WatchSpec(kind="Thing")], workers=4).start()
+    try:
+        store.create(_obj("a"))
+        store.create(_obj("b"))
+        # storm of updates on both keys while reconciles are in flight
+        for i in range(40):
+            for n in ("a", "b"):
+                o = store.get("Thing", n)
+                o["i"] = i
+                store.update(o)
+            time.sleep(0.002)
+        done.wait(timeout=10)
+        assert not overlaps, f"same key reconciled concurrently: {overlaps}"
+        assert calls[0] >= 4  # coalescing is allowed, starvation is not
+    finally:
+        eng.stop()
+
+
+def test_dirty_key_rereconciled_after_inflight():
+    """An event landing mid-reconcile must trigger one more pass (the state
+    the reconcile read may be stale)."""
+    store = MemStore()
+    started = threading.Event()
+    release = threading.Event()
+    passes = []
+
+    def rec(key):
+        passes.append(time.monotonic())
+        started.set()
+        release.wait(timeout=5)
+        return Result()
+
+    eng = Engine("t", store, rec, [WatchSpec(kind="Thing")], workers=1).start()
+    try:
+        store.create(_obj("a"))
+        assert started.wait(2)
+        o = store.get("Thing", "a")  # event arrives while reconcile blocked
+        o["x"] = 1
+        store.update(o)
+        time.sleep(0.05)
+        n_before = len(passes)
+        release.set()
+        deadline = time.monotonic() + 5
+        while len(passes) <= n_before and time.monotonic() < deadline:
+            time.sleep(0.01)
+        assert len(passes) > n_before, "dirty key was not re-reconciled"
+    finally:
+        release.set()
+        eng.stop()

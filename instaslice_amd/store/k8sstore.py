@@ -1,0 +1,265 @@
+"""Real Kubernetes API-server adapter with the MemStore verb set.
+
+On a cluster the controller/daemonset run against the API server itself; the
+reconcilers are store-agnostic, so this adapter is all that changes (the
+in-memory and TCP stores cover tests and benches). Requires the `kubernetes`
+pip package, which is intentionally NOT vendored — import is lazy and the
+constructor raises a clear error when it's missing (this build environment
+has no cluster to run against; tests/test_k8sstore.py covers the interface
+contract and skips the live part without the package).
+
+Mapping:
+  Pod / ConfigMap / Node  -> core v1 API
+  Instaslice              -> custom objects API (inference.codeflare.dev)
+  Lease                   -> coordination.k8s.io v1
+  resourceVersion conflicts -> kubernetes 409 -> Conflict
+  watch                   -> kubernetes.watch.Watch streaming thread
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+from typing import Callable, List, Optional, Tuple
+
+from instaslice_amd import API_GROUP, API_VERSION
+from instaslice_amd.store.memstore import AlreadyExists, Conflict, NotFound
+from instaslice_amd.utils import get_logger
+
+
+def _require_k8s():
+    try:
+        import kubernetes  # noqa: F401
+
+        return kubernetes
+    except ImportError as e:
+        raise ImportError(
+            "store.k8sstore requires the `kubernetes` package "
+            "(pip install kubernetes); use MemStore/NetStoreClient otherwise"
+        ) from e
+
+
+class _K8sWatch:
+    def __init__(self) -> None:
+        self._q: "queue.Queue[Optional[Tuple[str, dict]]]" = queue.Queue()
+        self._stopped = False
+        self.kind: Optional[str] = None
+
+    def next(self, timeout: Optional[float] = None):
+        try:
+            return self._q.get(timeout=timeout)
+        except queue.Empty:
+            return None
+
+    def stop(self) -> None:
+        self._stopped = True
+        self._q.put(None)
+
+
+class K8sStore:
+    """MemStore-compatible adapter over a live API server."""
+
+    def __init__(self, namespace_default: str = "default") -> None:
+        k8s = _require_k8s()
+        try:
+            k8s.config.load_incluster_config()
+        except Exception:
+            k8s.config.load_kube_config()
+        self._k8s = k8s
+        self._core = k8s.client.CoreV1Api()
+        self._custom = k8s.client.CustomObjectsApi()
+        self._coord = k8s.client.CoordinationV1Api()
+        self.namespace_default = namespace_default
+        self.log = get_logger("k8sstore")
+        self._watch_threads: List[threading.Thread] = []
+
+    # -- plumbing ------------------------------------------------------------
+
+    def _raise(self, e, kind, name):
+        status = getattr(e, "status", None)
+        if status == 404:
+            raise NotFound(f"{kind}/{name} not found") from None
+        if status == 409:
+            msg = str(getattr(e, "body", e))
+            if "AlreadyExists" in msg or "already exists" in msg:
+                raise AlreadyExists(f"{kind}/{name} exists") from None
+            raise Conflict(f"{kind}/{name}: {msg}") from None
+        raise
+
+    def _route(self, kind: str):
+        return kind  # dispatch happens in each verb
+
+    @staticmethod
+    def _to_dict(obj) -> dict:
+        if isinstance(obj, dict):
+            return obj
+        import kubernetes.client as client
+
+        return client.ApiClient().sanitize_for_serialization(obj)
+
+    # -- verbs ----------------------------------------------------------------
+
+    def create(self, obj: dict) -> dict:
+        kind = obj["kind"]
+        ns = obj["metadata"].get("namespace", self.namespace_default)
+        name = obj["metadata"]["name"]
+        ApiException = self._k8s.client.rest.ApiException
+        try:
+            if kind == "Instaslice":
+                return self._custom.create_namespaced_custom_object(
+                    API_GROUP, API_VERSION, ns, "instaslices", obj)
+            if kind == "Pod":
+                return self._to_dict(self._core.create_namespaced_pod(ns, obj))
+            if kind == "ConfigMap":
+                return self._to_dict(self._core.create_namespaced_config_map(ns, obj))
+            if kind == "Node":
+                return self._to_dict(self._core.create_node(obj))
+            if kind == "Lease":
+                return self._to_dict(self._coord.create_namespaced_lease(ns, obj))
+            raise ValueError(f"unsupported kind {kind}")
+        except ApiException as e:
+            self._raise(e, kind, name)
+
+    def get(self, kind: str, name: str, namespace: str = "") -> dict:
+        ns = namespace or self.namespace_default
+        ApiException = self._k8s.client.rest.ApiException
+        try:
+            if kind == "Instaslice":
+                return self._custom.get_namespaced_custom_object(
+                    API_GROUP, API_VERSION, ns, "instaslices", name)
+            if kind == "Pod":
+                return self._to_dict(self._core.read_namespaced_pod(name, ns))
+            if kind == "ConfigMap":
+                return self._to_dict(self._core.read_namespaced_config_map(name, ns))
+            if kind == "Node":
+                return self._to_dict(self._core.read_node(name))
+            if kind == "Lease":
+                return self._to_dict(self._coord.read_namespaced_lease(name, ns))
+            raise ValueError(f"unsupported kind {kind}")
+        except ApiException as e:
+            self._raise(e, kind, name)
+
+    def list(self, kind: str, namespace: Optional[str] = None) -> List[dict]:
+        ApiException = self._k8s.client.rest.ApiException
+        try:
+            if kind == "Instaslice":
+                res = self._custom.list_cluster_custom_object(
+                    API_GROUP, API_VERSION, "instaslices")
+                return list(res.get("items", []))
+            if kind == "Pod":
+                res = (self._core.list_namespaced_pod(namespace)
+                       if namespace else self._core.list_pod_for_all_namespaces())
+                return [self._to_dict(i) for i in res.items]
+            if kind == "ConfigMap":
+                res = (self._core.list_namespaced_config_map(namespace)
+                       if namespace
+                       else self._core.list_config_map_for_all_namespaces())
+                return [self._to_dict(i) for i in res.items]
+            if kind == "Node":
+                return [self._to_dict(i) for i in self._core.list_node().items]
+            raise ValueError(f"unsupported kind {kind}")
+        except ApiException as e:
+            self._raise(e, kind, "*")
+
+    def update(self, obj: dict) -> dict:
+        kind = obj["kind"]
+        ns = obj["metadata"].get("namespace", self.namespace_default)
+        name = obj["metadata"]["name"]
+        ApiException = self._k8s.client.rest.ApiException
+        try:
+            if kind == "Instaslice":
+                return self._custom.replace_namespaced_custom_object(
+                    API_GROUP, API_VERSION, ns, "instaslices", name, obj)
+            if kind == "Pod":
+                return self._to_dict(self._core.replace_namespaced_pod(name, ns, obj))
+            if kind == "ConfigMap":
+                return self._to_dict(
+                    self._core.replace_namespaced_config_map(name, ns, obj))
+            if kind == "Node":
+                # capacity patches go through the status subresource
+                return self._to_dict(self._core.patch_node_status(name, obj))
+            if kind == "Lease":
+                return self._to_dict(
+                    self._coord.replace_namespaced_lease(name, ns, obj))
+            raise ValueError(f"unsupported kind {kind}")
+        except ApiException as e:
+            self._raise(e, kind, name)
+
+    def delete(self, kind: str, name: str, namespace: str = "", *, now: float = 0.0) -> None:
+        ns = namespace or self.namespace_default
+        ApiException = self._k8s.client.rest.ApiException
+        try:
+            if kind == "Instaslice":
+                self._custom.delete_namespaced_custom_object(
+                    API_GROUP, API_VERSION, ns, "instaslices", name)
+            elif kind == "Pod":
+                self._core.delete_namespaced_pod(name, ns)
+            elif kind == "ConfigMap":
+                self._core.delete_namespaced_config_map(name, ns)
+            elif kind == "Node":
+                self._core.delete_node(name)
+            else:
+                raise ValueError(f"unsupported kind {kind}")
+        except ApiException as e:
+            self._raise(e, kind, name)
+
+    # -- watch ----------------------------------------------------------------
+
+    _WATCHABLE = {
+        "Pod": ("core", "list_pod_for_all_namespaces"),
+        "ConfigMap": ("core", "list_config_map_for_all_namespaces"),
+        "Node": ("core", "list_node"),
+    }
+
+    def watch(self, kind: Optional[str] = None, *, replay: bool = True):
+        if kind is None:
+            raise ValueError("K8sStore.watch requires an explicit kind")
+        w = _K8sWatch()
+        w.kind = kind
+
+        def pump():
+            kwatch = self._k8s.watch.Watch()
+            while not w._stopped:
+                try:
+                    if kind == "Instaslice":
+                        stream = kwatch.stream(
+                            self._custom.list_cluster_custom_object,
+                            API_GROUP, API_VERSION, "instaslices",
+                            timeout_seconds=30)
+                    else:
+                        api, fn = self._WATCHABLE[kind]
+                        target = getattr(
+                            self._core if api == "core" else self._coord, fn)
+                        stream = kwatch.stream(target, timeout_seconds=30)
+                    for ev in stream:
+                        if w._stopped:
+                            return
+                        obj = self._to_dict(ev["object"])
+                        w._q.put((ev["type"], obj))
+                except Exception as e:  # reconnect loop (API server restarts)
+                    if w._stopped:
+                        return
+                    self.log.warning("watch %s reconnecting: %s", kind, e)
+
+        t = threading.Thread(target=pump, daemon=True, name=f"k8swatch-{kind}")
+        t.start()
+        self._watch_threads.append(t)
+        return w
+
+    def update_with_retry(
+        self, kind: str, name: str, namespace: str,
+        mutate: Callable[[dict], Optional[dict]], attempts: int = 10,
+    ):
+        for _ in range(attempts):
+            try:
+                obj = self.get(kind, name, namespace)
+            except NotFound:
+                return None
+            new = mutate(obj)
+            if new is None:
+                return None
+            try:
+                return self.update(new)
+            except Conflict:
+                continue
+        raise Conflict(f"update_with_retry exhausted for {kind}/{name}")

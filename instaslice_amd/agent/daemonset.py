@@ -21,6 +21,7 @@ no custom metrics).
 
 from __future__ import annotations
 
+import threading
 import time
 from typing import Dict, List, Optional
 
@@ -55,6 +56,7 @@ class NodeAgent:
         smi: AmdSmi,
         node_name: str,
         reset_mode_on_empty: bool = False,
+        heartbeat_every_s: float = 5.0,
     ) -> None:
         self.store = store
         self.smi = smi
@@ -73,6 +75,12 @@ class NodeAgent:
         # consecutive hard mode-set failures per GPU (3 strikes -> fail alloc)
         self._mode_set_failures: Dict[str, int] = {}
         self.prepare_failures = 0  # observability counter
+        # failure-detection: periodic heartbeat into the CR status so the
+        # controller stops placing onto nodes whose agent died (the reference
+        # has no liveness signal at all)
+        self.heartbeat_every_s = heartbeat_every_s
+        self._hb_stop = threading.Event()
+        self._hb_thread = None
         self.engine = Engine(
             name=f"agent-{node_name}",
             store=store,
@@ -461,10 +469,34 @@ class NodeAgent:
 
     # -- lifecycle ------------------------------------------------------------
 
+    def _heartbeat_loop(self) -> None:
+        while not self._hb_stop.is_set():
+            def beat(obj: dict) -> Optional[dict]:
+                obj.setdefault("status", {})["heartbeat"] = time.time()
+                return obj
+
+            try:
+                self.store.update_with_retry(
+                    "Instaslice", self.node_name, INSTASLICE_NS, beat
+                )
+            except Exception as e:  # store outage: keep trying
+                self.log.warning("heartbeat failed: %s", e)
+            self._hb_stop.wait(self.heartbeat_every_s)
+
     def start(self) -> "NodeAgent":
         self.discover()
         self.engine.start()
+        if self.heartbeat_every_s > 0:
+            self._hb_stop.clear()
+            self._hb_thread = threading.Thread(
+                target=self._heartbeat_loop, daemon=True,
+                name=f"heartbeat-{self.node_name}",
+            )
+            self._hb_thread.start()
         return self
 
     def stop(self) -> None:
+        self._hb_stop.set()
+        if self._hb_thread:
+            self._hb_thread.join(timeout=2.0)
         self.engine.stop()

@@ -55,11 +55,17 @@ class PodController:
         policy: str = "packed-fit",
         teardown_grace_s: float = 30.0,
         requeue_no_fit_s: float = REQUEUE_NO_FIT_S,
+        node_stale_after_s: float = 30.0,
     ) -> None:
         self.store = store
         self.policy: AllocationPolicy = get_policy(policy)
         self.teardown_grace_s = teardown_grace_s
         self.requeue_no_fit_s = requeue_no_fit_s
+        # failure detection: nodes whose agent heartbeat is older than this
+        # receive no new placements, and their stuck `creating` allocations
+        # are re-placed elsewhere. 0 disables. (CRs without a heartbeat field
+        # — externally crafted — are treated as healthy.)
+        self.node_stale_after_s = node_stale_after_s
         self.log = get_logger("controller")
         # latency observability: pod_uid -> submit time, and measured p50 feed
         self.alloc_latency_s: List[float] = []
@@ -116,6 +122,12 @@ class PodController:
                 return p
         return None
 
+    def _node_stale(self, cr: dict) -> bool:
+        if not self.node_stale_after_s:
+            return False
+        hb = (cr.get("status") or {}).get("heartbeat")
+        return hb is not None and (time.time() - float(hb)) > self.node_stale_after_s
+
     def _try_place(self, pod: dict, profile_name: str, crs: List[dict]) -> Optional[Result]:
         """Scan nodes, place, persist the allocation (status=creating).
         Reference: findDeviceForASlice loop (instaslice_controller.go:192-222)."""
@@ -126,6 +138,8 @@ class PodController:
             node_name = cr["metadata"]["name"]
             if want_node and node_name != want_node:
                 continue
+            if self._node_stale(cr):
+                continue  # agent dead: no new placements here
             profile = self._profile_for(cr, profile_name)
             if profile is None:
                 continue  # node does not offer this profile
@@ -178,7 +192,11 @@ class PodController:
                     placement.gpu_uuid[:8], placement.ordinal,
                 )
                 self._mark_unschedulable(md, profile_name, unschedulable=False)
-                return Result()  # wait for agent's "created" event
+                # normally the agent's "created" event advances this pod; the
+                # timed recheck only notices a dead agent (no events then)
+                return Result(
+                    requeue_after=self.node_stale_after_s or None
+                )
         self._mark_unschedulable(md, profile_name, unschedulable=True)
         return Result(requeue_after=self.requeue_no_fit_s)
 
@@ -273,6 +291,26 @@ class PodController:
 
         cr, alloc = found
         status = alloc["allocationStatus"]
+        if status == AllocationStatus.CREATING and self._node_stale(cr):
+            # agent died before realizing it: reclaim and re-place elsewhere
+            # (running pods can't migrate — their partition is on that node —
+            # but pending ones should not wait for a dead agent)
+            def reclaim(cr_obj: dict) -> Optional[dict]:
+                allocs = cr_obj.get("spec", {}).get("allocations") or {}
+                a = allocs.get(uid)
+                if not a or a["allocationStatus"] != AllocationStatus.CREATING:
+                    return None
+                del allocs[uid]
+                return cr_obj
+
+            self.store.update_with_retry(
+                "Instaslice", cr["metadata"]["name"], INSTASLICE_NS, reclaim
+            )
+            self.log.warning(
+                "node %s stale; re-placing pending pod %s",
+                cr["metadata"]["name"], name,
+            )
+            return Result(requeue_after=0.01)
         if status == AllocationStatus.CREATED:
             # agent realized the partition: let the pod schedule
             def do_ungate(p: dict) -> Optional[dict]:
@@ -318,7 +356,11 @@ class PodController:
                 name, alloc["gpuUUID"][:8],
             )
             return Result(requeue_after=0.01)
-        # creating / ungated / deleted: nothing to do here
+        if status == AllocationStatus.CREATING and self.node_stale_after_s:
+            # agent normally advances this via events; the periodic recheck
+            # only exists to notice a dead agent (no events ever arrive then)
+            return Result(requeue_after=self.node_stale_after_s)
+        # ungated / deleted: nothing to do here
         return Result()
 
     # -- lifecycle ------------------------------------------------------------

@@ -2,6 +2,7 @@ from instaslice_amd.controller.policy import (  # noqa: F401
     AllocationPolicy,
     FirstFitPolicy,
     PackedFitPolicy,
+    SpreadFitPolicy,
     Placement,
     build_gpu_views,
     get_policy,

@@ -216,8 +216,39 @@ class PackedFitPolicy(AllocationPolicy):
         return best[1] if best else None
 
 
+class SpreadFitPolicy(AllocationPolicy):
+    """Bandwidth/thermal-headroom spreading: prefer the *least*-occupied GPU
+    already in the target mode, then idle GPUs. Co-tenants on one GPU share
+    its HBM (8 TB/s) and power budget; spreading maximizes per-pod headroom
+    at the cost of pinning more GPUs to a mode — the inverse trade of
+    PackedFitPolicy, exposed so operators can choose per cluster."""
+
+    name = "spread-fit"
+
+    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+        best: Optional[Tuple[Tuple, Placement]] = None
+        for v in views:
+            # occupancy is the primary key: an idle GPU (even one needing a
+            # mode flip) beats a loaded same-mode GPU — the opposite of
+            # packed-fit's tiering
+            if v.compute_mode is profile.compute:
+                p = _place_on(v, profile, needs_change=False)
+                if p is None:
+                    continue
+                score = (-len(v.occupied), 1, -v.index)  # tie: prefer no flip
+            else:
+                p = _place_on(v, profile, needs_change=True)
+                if p is None:
+                    continue
+                mem_ok = v.memory_mode in VALID_MEMORY_MODES[profile.compute]
+                score = (0, 0, (1 if mem_ok else 0) - v.index)
+            if best is None or score > best[0]:
+                best = (score, p)
+        return best[1] if best else None
+
+
 POLICIES: Dict[str, AllocationPolicy] = {
-    p.name: p for p in (FirstFitPolicy(), PackedFitPolicy())
+    p.name: p for p in (FirstFitPolicy(), PackedFitPolicy(), SpreadFitPolicy())
 }
 
 

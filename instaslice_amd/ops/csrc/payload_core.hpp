@@ -72,6 +72,23 @@ __global__ __launch_bounds__(kBlock) void stream_copy_kernel(
   for (; i < n4; i += stride) dst[i] = src[i];
 }
 
+// Nontemporal variant: streamed data is used exactly once, so bypassing the
+// L2/LLC write-allocate path frees cache bandwidth for the reads.
+__global__ __launch_bounds__(kBlock) void stream_copy_nt_kernel(
+    const float4* __restrict__ src, float4* __restrict__ dst, size_t n4) {
+  size_t i = blockIdx.x * static_cast<size_t>(blockDim.x) + threadIdx.x;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  // the builtin wants a scalar/vector-of-scalar pointer, not the
+  // HIP_vector_type struct: reinterpret as the native 4-float vector
+  typedef float vfloat4 __attribute__((ext_vector_type(4)));
+  const vfloat4* s = reinterpret_cast<const vfloat4*>(src);
+  vfloat4* d = reinterpret_cast<vfloat4*>(dst);
+  for (; i < n4; i += stride) {
+    vfloat4 v = __builtin_nontemporal_load(&s[i]);
+    __builtin_nontemporal_store(v, &d[i]);
+  }
+}
+
 // Bounded busy-wait on the constant-rate wall clock (s_memrealtime). The
 // iteration guard bounds the spin even if the clock misbehaves
 // (cdna_hip_programming.md §1: "bound every spin").
@@ -147,24 +164,29 @@ inline double run_vecadd(size_t n, int device = 0) {
 }
 
 // Streaming-copy bandwidth in GB/s (read+write bytes counted).
-inline double run_membw(size_t bytes, int iters, int device = 0) {
+// blocks=0 picks the default grid; pass an explicit count to sweep the
+// workgroups-per-CU space (MI355X: 256 CUs x 8 XCDs; the guide's rule is
+// >> 256 workgroups to fill the chip).
+inline double run_membw(size_t bytes, int iters, int device = 0, int blocks = 0,
+                        bool nontemporal = false) {
   HIP_CHECK(hipSetDevice(device));
   size_t n4 = bytes / sizeof(float4);
   if (n4 == 0) throw std::invalid_argument("membw: bytes too small");
   float4 *src, *dst;
   HIP_CHECK(hipMalloc(&src, n4 * sizeof(float4)));
   HIP_CHECK(hipMalloc(&dst, n4 * sizeof(float4)));
-  int grid = grid_for(n4);
+  int grid = blocks > 0 ? blocks : grid_for(n4);
+  auto kernel = nontemporal ? stream_copy_nt_kernel : stream_copy_kernel;
   hipLaunchKernelGGL(fill_kernel, dim3(grid), dim3(kBlock), 0, 0, src, 1.0f, n4);
   // warmup
-  hipLaunchKernelGGL(stream_copy_kernel, dim3(grid), dim3(kBlock), 0, 0, src, dst, n4);
+  hipLaunchKernelGGL(kernel, dim3(grid), dim3(kBlock), 0, 0, src, dst, n4);
   HIP_CHECK(hipDeviceSynchronize());
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
   HIP_CHECK(hipEventRecord(t0, 0));
   for (int i = 0; i < iters; ++i) {
-    hipLaunchKernelGGL(stream_copy_kernel, dim3(grid), dim3(kBlock), 0, 0, src, dst, n4);
+    hipLaunchKernelGGL(kernel, dim3(grid), dim3(kBlock), 0, 0, src, dst, n4);
   }
   HIP_CHECK(hipEventRecord(t1, 0));
   HIP_CHECK(hipEventSynchronize(t1));

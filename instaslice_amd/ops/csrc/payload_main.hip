@@ -46,11 +46,13 @@ int main(int argc, char** argv) {
     } else if (std::strcmp(cmd, "membw") == 0) {
       size_t bytes = argc > 2 ? std::strtoull(argv[2], nullptr, 10) : (size_t{1} << 30);
       int iters = argc > 3 ? std::atoi(argv[3]) : 10;
-      double gbs = run_membw(bytes, iters);
+      int blocks = argc > 4 ? std::atoi(argv[4]) : 0;
+      bool nt = argc > 5 && std::atoi(argv[5]) != 0;
+      double gbs = run_membw(bytes, iters, 0, blocks, nt);
       std::printf(
           "{\"ok\": true, \"cmd\": \"membw\", \"bytes\": %zu, \"iters\": %d, "
-          "\"gb_per_s\": %.1f}\n",
-          bytes, iters, gbs);
+          "\"blocks\": %d, \"nontemporal\": %s, \"gb_per_s\": %.1f}\n",
+          bytes, iters, blocks, nt ? "true" : "false", gbs);
     } else if (std::strcmp(cmd, "busy") == 0) {
       double ms = argc > 2 ? std::atof(argv[2]) : 100.0;
       run_busy(ms);

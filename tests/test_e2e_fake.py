@@ -258,3 +258,33 @@ def test_mode_locked_gpu_replacement():
         c.wait_pod_scheduled("p2", timeout=15.0)
     finally:
         c.stop()
+
+
+def test_concurrent_submissions_race_free():
+    """20 pods submitted from 4 threads simultaneously: every pod schedules,
+    every (gpu, ordinal) is unique — the placement race the reference's
+    unlocked caches could lose (SURVEY.md §5 race-detection note)."""
+    import threading
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=4, node_name="node-0"))
+    c.start()
+    try:
+        def submit(base):
+            for i in range(5):
+                c.submit_pod(f"r{base}-{i}", "cpx-1x36")
+
+        threads = [threading.Thread(target=submit, args=(t,)) for t in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        for t in range(4):
+            for i in range(5):
+                c.wait_pod_scheduled(f"r{t}-{i}", timeout=20.0)
+        prepared = c.prepared("node-0")
+        assert len(prepared) == 20
+        slots = {(p["parentGpuUUID"], p["ordinal"]) for p in prepared.values()}
+        assert len(slots) == 20, "duplicate (gpu, ordinal) handed out"
+    finally:
+        c.stop()

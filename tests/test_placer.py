@@ -141,3 +141,27 @@ def test_packedfit_leaves_room_for_spx():
     assert views[0].compute_mode is ComputeMode.CPX and len(views[0].occupied) == 7
     p = policy.place(SPX, views)
     assert p is not None and p.gpu_uuid == "gpu-1"
+
+
+def test_spreadfit_balances_across_gpus():
+    from instaslice_amd.controller.policy import SpreadFitPolicy
+
+    views = [mk_view(0), mk_view(1)]
+    policy = SpreadFitPolicy()
+    parents = []
+    for _ in range(4):
+        p = policy.place(CPX, views)
+        v = next(v for v in views if v.uuid == p.gpu_uuid)
+        if p.needs_mode_change:
+            v.compute_mode = ComputeMode.CPX
+        v.occupied.add(p.ordinal)
+        parents.append(p.gpu_uuid)
+    # 4 pods over 2 GPUs: 2 each (packed-fit would put all 4 on gpu-0)
+    assert parents.count("gpu-0") == 2 and parents.count("gpu-1") == 2
+
+
+@pytest.mark.parametrize("policy_name", ["first-fit", "packed-fit", "spread-fit"])
+def test_policy_registry(policy_name):
+    from instaslice_amd.controller.policy import get_policy
+
+    assert get_policy(policy_name).name == policy_name

@@ -49,12 +49,17 @@ def _wait_forever():
 
 
 def cmd_store(args) -> int:
+    from instaslice_amd.store.memstore import MemStore
     from instaslice_amd.store.netstore import StoreServer
 
-    server = StoreServer(port=args.port).start()
-    log.info("store serving on %s:%d", server.host, server.port)
+    backing = MemStore(persist_path=args.data) if args.data else None
+    server = StoreServer(store=backing, port=args.port).start()
+    log.info("store serving on %s:%d (data=%s)", server.host, server.port,
+             args.data or "in-memory")
     _wait_forever()
     server.stop()
+    if backing:
+        backing.close()
     return 0
 
 
@@ -192,6 +197,8 @@ def main(argv=None) -> int:
 
     p = sub.add_parser("store", help="run the state store server")
     p.add_argument("--port", type=int, default=7080)
+    p.add_argument("--data", default=None,
+                   help="persist state to this JSON file (checkpoint/resume)")
     p.set_defaults(fn=cmd_store)
 
     p = sub.add_parser("controller", help="run the cluster controller")

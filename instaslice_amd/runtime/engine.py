@@ -23,6 +23,19 @@ from instaslice_amd.utils import get_logger
 
 Key = Tuple[str, str, str]  # (kind, namespace, name)
 
+# process-wide reconcile trace (bounded ring; see metrics.Tracer). Dumped via
+# `python -m instaslice_amd status` debugging or read directly in tests.
+_tracer = None
+
+
+def get_tracer():
+    global _tracer
+    if _tracer is None:
+        from instaslice_amd.metrics import Tracer
+
+        _tracer = Tracer(capacity=8192)
+    return _tracer
+
 
 @dataclass
 class Result:
@@ -148,8 +161,12 @@ class Engine:
                 self.enqueue_after(key, self.error_backoff_s)
                 continue
             requeued = res and res.requeue_after is not None
-            metrics.reconcile(self.name, "requeue" if requeued else "ok",
-                              time.monotonic() - t0)
+            dt = time.monotonic() - t0
+            metrics.reconcile(self.name, "requeue" if requeued else "ok", dt)
+            get_tracer().event(
+                "reconcile", engine=self.name, key=list(key),
+                duration_s=round(dt, 6), requeued=bool(requeued),
+            )
             self._finish(key)
             if requeued:
                 self.enqueue_after(key, res.requeue_after)

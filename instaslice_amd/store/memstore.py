@@ -90,11 +90,90 @@ class MemStore:
     store (matching client-go cache semantics the reference relies on).
     """
 
-    def __init__(self) -> None:
+    def __init__(self, persist_path: Optional[str] = None,
+                 persist_debounce_s: float = 0.2) -> None:
         self._lock = threading.RLock()
         self._objects: Dict[Key, dict] = {}
         self._rv = 0
         self._watches: List[Watch] = []
+        # Durability (checkpoint/resume): the reference keeps all state in
+        # etcd (SURVEY.md §5); for standalone deployments this store can
+        # snapshot itself to a JSON file (write-behind, debounced) and reload
+        # it at boot, so controller/agent restarts adopt the same state.
+        self._persist_path = persist_path
+        self._persist_debounce_s = persist_debounce_s
+        self._persist_pending = threading.Event()
+        self._persist_thread: Optional[threading.Thread] = None
+        self._persist_stop = threading.Event()
+        if persist_path:
+            self._load()
+            self._persist_thread = threading.Thread(
+                target=self._persist_loop, daemon=True, name="store-persist"
+            )
+            self._persist_thread.start()
+
+    # -- persistence --------------------------------------------------------
+
+    def _load(self) -> None:
+        import os
+
+        if not self._persist_path or not os.path.exists(self._persist_path):
+            return
+        with open(self._persist_path) as f:
+            snap = json.load(f)
+        with self._lock:
+            self._rv = snap.get("rv", 0)
+            for obj in snap.get("objects", []):
+                self._objects[_key(obj)] = obj
+
+    def _persist_loop(self) -> None:
+        import os
+        import tempfile
+
+        while not self._persist_stop.is_set():
+            if not self._persist_pending.wait(timeout=0.5):
+                continue
+            # debounce: batch bursts of mutations into one write
+            self._persist_stop.wait(self._persist_debounce_s)
+            self._persist_pending.clear()
+            with self._lock:
+                snap = {"rv": self._rv, "objects": list(self._objects.values())}
+            d = os.path.dirname(os.path.abspath(self._persist_path)) or "."
+            fd, tmp = tempfile.mkstemp(dir=d, prefix=".store-")
+            try:
+                with os.fdopen(fd, "w") as f:
+                    json.dump(snap, f)
+                os.replace(tmp, self._persist_path)  # atomic
+            except OSError:
+                try:
+                    os.unlink(tmp)
+                except OSError:
+                    pass
+
+    def flush(self) -> None:
+        """Force one synchronous snapshot (tests / clean shutdown)."""
+        if not self._persist_path:
+            return
+        import os
+        import tempfile
+
+        with self._lock:
+            snap = {"rv": self._rv, "objects": list(self._objects.values())}
+        d = os.path.dirname(os.path.abspath(self._persist_path)) or "."
+        fd, tmp = tempfile.mkstemp(dir=d, prefix=".store-")
+        with os.fdopen(fd, "w") as f:
+            json.dump(snap, f)
+        os.replace(tmp, self._persist_path)
+
+    def close(self) -> None:
+        self._persist_stop.set()
+        if self._persist_thread:
+            self._persist_thread.join(timeout=2.0)
+        self.flush()
+
+    def _mark_dirty(self) -> None:
+        if self._persist_path:
+            self._persist_pending.set()
 
     # -- verbs ------------------------------------------------------------
 
@@ -108,6 +187,7 @@ class MemStore:
             obj.setdefault("metadata", {})["resourceVersion"] = str(self._rv)
             self._objects[k] = obj
             self._notify("ADDED", obj)
+            self._mark_dirty()
             return _snapshot(obj)
 
     def get(self, kind: str, name: str, namespace: str = "") -> dict:
@@ -154,6 +234,7 @@ class MemStore:
                 self._notify("DELETED", obj)
             else:
                 self._notify("MODIFIED", obj)
+            self._mark_dirty()
             return _snapshot(obj)
 
     def delete(self, kind: str, name: str, namespace: str = "", *, now: float = 0.0) -> None:
@@ -175,6 +256,7 @@ class MemStore:
             else:
                 del self._objects[k]
                 self._notify("DELETED", obj)
+            self._mark_dirty()
 
     # -- watch ------------------------------------------------------------
 

@@ -110,3 +110,51 @@ def test_update_with_retry_absorbs_conflicts():
     for t in ts:
         t.join()
     assert s.get("Thing", "a")["n"] == n_threads * n_incr
+
+
+def test_persistence_roundtrip(tmp_path):
+    """Checkpoint/resume: a store restart reloads all objects + rv (the
+    durability the reference gets from etcd)."""
+    path = str(tmp_path / "state.json")
+    s1 = MemStore(persist_path=path)
+    s1.create(_obj("a", n=1))
+    s1.create(new_pod("p1", profile="cpx-1x36"))
+    o = s1.get("Thing", "a")
+    o["n"] = 2
+    s1.update(o)
+    s1.close()
+
+    s2 = MemStore(persist_path=path)
+    assert s2.get("Thing", "a")["n"] == 2
+    pod = s2.get("Pod", "p1", "default")
+    assert pod["metadata"]["finalizers"]
+    # resourceVersion survives: no rv reuse after restart
+    o2 = s2.get("Thing", "a")
+    o2["n"] = 3
+    s2.update(o2)
+    assert int(s2.get("Thing", "a")["metadata"]["resourceVersion"]) > int(
+        o["metadata"]["resourceVersion"])
+    s2.close()
+
+
+def test_persistence_debounced_write_behind(tmp_path):
+    import json as _json
+    import time as _time
+
+    path = str(tmp_path / "state.json")
+    s = MemStore(persist_path=path, persist_debounce_s=0.05)
+    for i in range(20):
+        s.create(_obj(f"x{i}"))
+    deadline = _time.monotonic() + 5
+    while _time.monotonic() < deadline:
+        try:
+            with open(path) as f:
+                snap = _json.load(f)
+            if len(snap["objects"]) == 20:
+                break
+        except (OSError, ValueError):
+            pass
+        _time.sleep(0.02)
+    else:
+        raise AssertionError("write-behind never flushed all objects")
+    s.close()

@@ -23,6 +23,9 @@ bench:   ## flagship benchmark, single GPU
 scenarios:  ## BASELINE.json measurement configs 1-5 (fake-SMI)
 	$(PYTHON) -m benchmarks.scenarios
 
+lint:    ## static checks (ruff if present, else a strict compile pass)
+	@if command -v ruff >/dev/null 2>&1; then 	    ruff check instaslice_amd tests benchmarks bench.py; 	else 	    $(PYTHON) -m compileall -q instaslice_amd tests benchmarks bench.py 	        __graft_entry__.py build_native.py && echo "compileall OK (ruff unavailable)"; 	fi
+
 crd:     ## regenerate the CRD manifest from api/crd.py
 	$(PYTHON) -m instaslice_amd.api.crd > \
 	    config/crd/bases/inference.codeflare.dev_instaslices.yaml

@@ -216,7 +216,8 @@ def main() -> int:
     if rank == 0:
         server = StoreServer().start()
         store0 = server.store
-        controller = PodController(store0, policy=args.policy, teardown_grace_s=0.0)
+        controller = PodController(store0, policy=args.policy, teardown_grace_s=0.0,
+                                   workers=4)
         controller.requeue_no_fit_s = 0.05
         controller.start()
         addr = ("127.0.0.1", server.port)

@@ -56,6 +56,7 @@ class PodController:
         teardown_grace_s: float = 30.0,
         requeue_no_fit_s: float = REQUEUE_NO_FIT_S,
         node_stale_after_s: float = 30.0,
+        workers: int = 2,
     ) -> None:
         self.store = store
         self.policy: AllocationPolicy = get_policy(policy)
@@ -73,6 +74,10 @@ class PodController:
         # pods currently marked unschedulable: re-reconciled on ANY node-state
         # change so freed capacity is picked up event-driven, not by polling
         self._unschedulable_keys: set = set()
+        # workers > 1 is safe: the engine guarantees a key is never
+        # reconciled concurrently (runtime/engine.py in-flight guard), and
+        # placement races between different pods are caught by the
+        # re-validation inside add_alloc + resourceVersion conflicts
         self.engine = Engine(
             name="controller",
             store=store,
@@ -81,6 +86,7 @@ class PodController:
                 WatchSpec(kind="Pod"),
                 WatchSpec(kind="Instaslice", map_fn=self._instaslice_to_pods),
             ],
+            workers=workers,
         )
 
     # -- watch mapping ------------------------------------------------------

@@ -95,8 +95,11 @@ class SingleGpuSmi(AmdSmi):
 
 def make_smi(args, rank: int):
     """Real amd-smi if GPUs enumerate (no HIP context involved), else the
-    fake 1x MI355X model."""
+    fake 1x MI355X model. On a machine WITH a GPU (/dev/kfd present) the
+    native path is mandatory — a missing extension fails loudly instead of
+    silently measuring the fake."""
     if not args.fake:
+        gpu_box = os.path.exists("/dev/kfd")
         try:
             from instaslice_amd.smi.native import NativeAmdSmi
 
@@ -106,8 +109,15 @@ def make_smi(args, rank: int):
             if gpus:
                 local = int(os.environ.get("LOCAL_RANK", rank))
                 return SingleGpuSmi(smi, min(local, len(gpus) - 1)), "amdsmi"
+            if gpu_box:
+                raise SmiError("/dev/kfd present but amdsmi enumerated no GPUs")
             log.warning("amdsmi enumerated no GPUs; falling back to fake")
         except (SmiError, ImportError) as e:
+            if gpu_box:
+                raise RuntimeError(
+                    f"GPU present but native device layer unusable: {e} "
+                    "(run build_native.py; refusing fake fallback)"
+                ) from e
             log.warning("native smi unavailable (%s); falling back to fake", e)
     return FakeAmdSmi(num_gpus=1, node_name=f"node-{rank}"), "fake"
 

@@ -244,8 +244,35 @@ def scenario_churn(args) -> dict:
     }
 
 
+
+def scenario_flatlat(args) -> dict:
+    """Allocation latency vs live-partition count: the north-star requires
+    latency to stay FLAT as partition count grows (cached enumeration; the
+    reference re-inits NVML per reconcile and degrades). Fill 8 GPUs with 64
+    CPX pods, bucketing latency by load level."""
+    c = _mk_cluster(num_gpus=8)
+    buckets = {}
+    try:
+        for i in range(64):
+            t0 = time.perf_counter()
+            c.submit_pod(f"fl-{i}", "cpx-1x36")
+            c.wait_pod_scheduled(f"fl-{i}", timeout=15.0)
+            dt = (time.perf_counter() - t0) * 1000
+            buckets.setdefault(i // 16, []).append(dt)  # 0-15, 16-31, ...
+    finally:
+        c.stop()
+    out = {"scenario": "flatlat",
+           "p50_by_load": {f"pods_{16*k}_{16*k+15}":
+                           round(statistics.median(v), 3)
+                           for k, v in sorted(buckets.items())}}
+    p50s = list(out["p50_by_load"].values())
+    out["flatness_max_over_min"] = round(max(p50s) / min(p50s), 2)
+    return out
+
+
 SCENARIOS = {
     "plumbing": scenario_plumbing,
+    "flatlat": scenario_flatlat,
     "cpx8": scenario_cpx8,
     "mixed100": scenario_mixed100,
     "vllm": scenario_vllm,
@@ -272,3 +299,4 @@ def main(argv=None) -> int:
 
 if __name__ == "__main__":
     sys.exit(main())
+

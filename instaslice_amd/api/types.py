@@ -64,6 +64,7 @@ class AllocationDetails:
     allocation_status: str = AllocationStatus.CREATING
     compute_mode: str = ""       # target compute mode, e.g. "CPX"
     memory_mode: str = ""        # target memory mode, e.g. "NPS4"
+    group: str = ""              # gang label (org.instaslice/group annotation)
 
     def to_dict(self) -> dict:
         return {
@@ -79,6 +80,7 @@ class AllocationDetails:
             "allocationStatus": self.allocation_status,
             "computeMode": self.compute_mode,
             "memoryMode": self.memory_mode,
+            "group": self.group,
         }
 
     @classmethod
@@ -96,6 +98,7 @@ class AllocationDetails:
             allocation_status=d.get("allocationStatus", AllocationStatus.CREATING),
             compute_mode=d.get("computeMode", ""),
             memory_mode=d.get("memoryMode", ""),
+            group=d.get("group", ""),
         )
 
 
@@ -211,6 +214,7 @@ def new_pod(
     uid: Optional[str] = None,
     node_selector: Optional[Dict[str, str]] = None,
     gated: bool = True,
+    group: Optional[str] = None,
 ) -> dict:
     """Synthetic gated pod following the reference's consumer contract
     (samples/test-pod.yaml:1-21): scheduling gate + finalizer + profile limit
@@ -231,6 +235,9 @@ def new_pod(
             "uid": uid or new_uid(),
             "finalizers": [FINALIZER_NAME],
             "deletionTimestamp": None,
+            "annotations": (
+                {"org.instaslice/group": group} if group else {}
+            ),
         },
         "spec": {
             "schedulingGates": [{"name": GATE_NAME}] if gated else [],

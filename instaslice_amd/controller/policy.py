@@ -128,12 +128,27 @@ def _target_memory_mode(profile: PartitionProfile, current: MemoryMode) -> Tuple
 
 
 class AllocationPolicy:
-    """Strategy interface (reference: instaslice_controller.go:48-50)."""
+    """Strategy interface (reference: instaslice_controller.go:48-50).
+
+    `prefer_gpus` carries gang-placement affinity: GPUs already hosting the
+    pod's group. Same-GPU XCD co-location is free bandwidth (intra-die,
+    no xGMI hop — SURVEY.md §5), so preferred same-mode fits win over
+    everything else in every policy."""
 
     name = "base"
 
-    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+    def place(self, profile: PartitionProfile, views: List[GpuView],
+              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
         raise NotImplementedError
+
+    def _preferred(self, profile: PartitionProfile, views: List[GpuView],
+                   prefer_gpus: frozenset) -> Optional[Placement]:
+        for v in views:
+            if v.uuid in prefer_gpus and v.compute_mode is profile.compute:
+                p = _place_on(v, profile, needs_change=False)
+                if p:
+                    return p
+        return None
 
 
 def _free_ordinal(view: GpuView, n_partitions: int) -> Optional[int]:
@@ -165,7 +180,12 @@ class FirstFitPolicy(AllocationPolicy):
 
     name = "first-fit"
 
-    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+    def place(self, profile: PartitionProfile, views: List[GpuView],
+              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
+        if prefer_gpus:
+            p = self._preferred(profile, views, prefer_gpus)
+            if p:
+                return p
         for v in views:
             if v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
@@ -196,7 +216,12 @@ class PackedFitPolicy(AllocationPolicy):
 
     name = "packed-fit"
 
-    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+    def place(self, profile: PartitionProfile, views: List[GpuView],
+              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
+        if prefer_gpus:
+            p = self._preferred(profile, views, prefer_gpus)
+            if p:
+                return p
         best: Optional[Tuple[Tuple, Placement]] = None
         for v in views:
             if v.compute_mode is profile.compute:
@@ -225,7 +250,12 @@ class SpreadFitPolicy(AllocationPolicy):
 
     name = "spread-fit"
 
-    def place(self, profile: PartitionProfile, views: List[GpuView]) -> Optional[Placement]:
+    def place(self, profile: PartitionProfile, views: List[GpuView],
+              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
+        if prefer_gpus:
+            p = self._preferred(profile, views, prefer_gpus)
+            if p:
+                return p
         best: Optional[Tuple[Tuple, Placement]] = None
         for v in views:
             # occupancy is the primary key: an idle GPU (even one needing a

@@ -140,6 +140,7 @@ class PodController:
         md = pod["metadata"]
         node_selector = pod.get("spec", {}).get("nodeSelector") or {}
         want_node = node_selector.get("kubernetes.io/hostname")
+        group = (md.get("annotations") or {}).get("org.instaslice/group", "")
         for cr in crs:
             node_name = cr["metadata"]["name"]
             if want_node and node_name != want_node:
@@ -150,7 +151,14 @@ class PodController:
             if profile is None:
                 continue  # node does not offer this profile
             views = build_gpu_views(node_name, cr.get("spec", {}))
-            placement = self.policy.place(profile, views)
+            # gang affinity: GPUs already hosting this pod's group on this
+            # node (same-GPU XCD co-location = intra-die bandwidth, no xGMI)
+            prefer = frozenset(
+                a["gpuUUID"]
+                for a in (cr["spec"].get("allocations") or {}).values()
+                if group and a.get("group") == group
+            )
+            placement = self.policy.place(profile, views, prefer_gpus=prefer)
             if placement is None:
                 continue
             alloc = AllocationDetails(
@@ -166,6 +174,7 @@ class PodController:
                 allocation_status=AllocationStatus.CREATING,
                 compute_mode=placement.compute_mode,
                 memory_mode=placement.memory_mode,
+                group=group,
             )
 
             def add_alloc(cr_obj: dict, a=alloc) -> Optional[dict]:

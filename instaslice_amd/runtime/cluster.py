@@ -61,10 +61,11 @@ class Cluster:
 
     def submit_pod(
         self, name: str, profile: str, namespace: str = "default",
-        node: Optional[str] = None,
+        node: Optional[str] = None, group: Optional[str] = None,
     ) -> dict:
         sel = {"kubernetes.io/hostname": node} if node else None
-        pod = new_pod(name, namespace=namespace, profile=profile, node_selector=sel)
+        pod = new_pod(name, namespace=namespace, profile=profile,
+                      node_selector=sel, group=group)
         return self.store.create(pod)
 
     def wait_pod_scheduled(self, name: str, namespace: str = "default",

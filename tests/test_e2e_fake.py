@@ -288,3 +288,46 @@ def test_concurrent_submissions_race_free():
         assert len(slots) == 20, "duplicate (gpu, ordinal) handed out"
     finally:
         c.stop()
+
+
+def test_gang_placement_colocates_group():
+    """Pods sharing org.instaslice/group land on the SAME physical GPU when
+    capacity allows: same-GPU XCDs talk intra-die, not over xGMI
+    (SURVEY.md §5 co-placement note)."""
+    c = Cluster(teardown_grace_s=0.0, policy="spread-fit")
+    c.add_node("node-0", FakeAmdSmi(num_gpus=4, node_name="node-0"))
+    c.start()
+    try:
+        # spread-fit would normally scatter these across GPUs; the gang
+        # affinity must override that
+        for i in range(4):
+            c.submit_pod(f"g{i}", "cpx-1x36", group="trainer")
+            c.wait_pod_scheduled(f"g{i}")
+        prepared = c.prepared("node-0")
+        parents = {p["parentGpuUUID"] for p in prepared.values()}
+        assert len(parents) == 1, f"gang split across {len(parents)} GPUs"
+        # an ungrouped pod still follows the base policy (different GPU)
+        c.submit_pod("solo", "cpx-1x36")
+        c.wait_pod_scheduled("solo")
+        prepared = c.prepared("node-0")
+        assert len({p["parentGpuUUID"] for p in prepared.values()}) == 2
+    finally:
+        c.stop()
+
+
+def test_gang_spills_when_gpu_full():
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=2, node_name="node-0"))
+    c.start()
+    try:
+        for i in range(9):  # 9 members > 8 CPX slots per GPU
+            c.submit_pod(f"g{i}", "cpx-1x36", group="big")
+            c.wait_pod_scheduled(f"g{i}")
+        prepared = c.prepared("node-0")
+        by_parent = {}
+        for p in prepared.values():
+            by_parent.setdefault(p["parentGpuUUID"], []).append(p)
+        sizes = sorted(len(v) for v in by_parent.values())
+        assert sizes == [1, 8]  # one full GPU + one spill
+    finally:
+        c.stop()

@@ -42,13 +42,13 @@ import torch
 import torch.distributed as dist
 
 from instaslice_amd.agent.daemonset import NodeAgent
-from instaslice_amd.controller.reconciler import PodController
 from instaslice_amd.partition.profiles import ProfileCatalog
 from instaslice_amd.runtime.cluster import Cluster  # noqa: F401 (doc anchor)
+from instaslice_amd.runtime.controlplane import run_control_plane
 from instaslice_amd.smi.base import AmdSmi, SmiBusy, SmiError
 from instaslice_amd.smi.fake import FakeAmdSmi
-from instaslice_amd.store.memstore import MemStore, NotFound
-from instaslice_amd.store.netstore import NetStoreClient, StoreServer
+from instaslice_amd.store.memstore import NotFound
+from instaslice_amd.store.netstore import NetStoreClient
 from instaslice_amd.utils import get_logger
 
 log = get_logger("bench")
@@ -211,6 +211,10 @@ def main() -> int:
     ap.add_argument("--seed", type=int, default=1234)
     args = ap.parse_args()
 
+    # the lifecycle chain is thread-wakeup bound; don't let a CPU-holding
+    # thread keep the GIL for the default 5 ms
+    sys.setswitchinterval(0.001)
+
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
 
@@ -221,16 +225,21 @@ def main() -> int:
         # child processes inside partitions.
         dist.init_process_group(backend="gloo")
 
-    # -- topology: rank 0 = store server + controller; all ranks = 1 agent --
-    server = controller = None
+    # -- topology: rank 0 spawns the control-plane PROCESS (store server +
+    # controller, its own GIL — as in production, where the controller is its
+    # own pod); all ranks (incl. 0) run one node agent over TCP --
+    cp_proc = cp_conn = None
     if rank == 0:
-        server = StoreServer().start()
-        store0 = server.store
-        controller = PodController(store0, policy=args.policy, teardown_grace_s=0.0,
-                                   workers=4)
-        controller.requeue_no_fit_s = 0.05
-        controller.start()
-        addr = ("127.0.0.1", server.port)
+        import multiprocessing as mp
+
+        ctx = mp.get_context("spawn")  # no torch/threads inherited
+        cp_conn, child_conn = ctx.Pipe()
+        cp_proc = ctx.Process(
+            target=run_control_plane, args=(child_conn, args.policy),
+            daemon=True, name="control-plane",
+        )
+        cp_proc.start()
+        addr = ("127.0.0.1", cp_conn.recv())
     else:
         addr = None
     if world > 1:
@@ -265,8 +274,16 @@ def main() -> int:
             dist.barrier()
 
     # -- one step = one pod lifecycle, waits driven by watch events ----------
+    # filtered subscription: only THIS rank's pods (by label) and THIS rank's
+    # node CR cross the wire. An unfiltered all-kinds watch on every rank is
+    # O(ranks^2) event traffic through the store — measured as the biggest
+    # multi-rank scaling limiter.
     latencies_ms = []
-    events = store.watch(None, replay=False)  # all kinds this rank touches
+    bench_label = {"org.instaslice/bench-rank": str(rank)}
+    events = store.watch(replay=False, filters=[
+        {"kind": "Pod", "labels": bench_label},
+        {"kind": "Instaslice", "name": node},
+    ])
 
     def wait_event(pred, what: str, timeout: float = 120.0) -> None:
         deadline = time.monotonic() + timeout
@@ -287,7 +304,8 @@ def main() -> int:
 
         t0 = time.perf_counter()
         store.create(new_pod(name, profile=prof,
-                             node_selector={"kubernetes.io/hostname": node}))
+                             node_selector={"kubernetes.io/hostname": node},
+                             labels=bench_label))
 
         def scheduled(et, obj):
             return (obj["kind"] == "Pod"
@@ -382,11 +400,15 @@ def main() -> int:
     # teardown
     events.stop()
     agent.stop()
-    if controller:
-        controller.stop()
     store.close()
-    if server:
-        server.stop()
+    if cp_proc is not None:
+        try:
+            cp_conn.send("stop")
+        except (BrokenPipeError, OSError):
+            pass
+        cp_proc.join(timeout=5.0)
+        if cp_proc.is_alive():
+            cp_proc.terminate()
     if world > 1:
         dist.destroy_process_group()
     return 0

@@ -81,11 +81,17 @@ class NodeAgent:
         self.heartbeat_every_s = heartbeat_every_s
         self._hb_stop = threading.Event()
         self._hb_thread = None
+        # filtered watch: only THIS node's CR crosses the wire (a cluster
+        # with N agents otherwise broadcasts every CR event to all N)
         self.engine = Engine(
             name=f"agent-{node_name}",
             store=store,
             reconcile=self._reconcile,
-            watches=[WatchSpec(kind="Instaslice", map_fn=self._own_node_only)],
+            watches=[WatchSpec(
+                kind="Instaslice",
+                map_fn=self._own_node_only,
+                filters=[{"kind": "Instaslice", "name": node_name}],
+            )],
         )
 
     def _own_node_only(self, event_type: str, obj: dict) -> List[Key]:
@@ -229,8 +235,10 @@ class NodeAgent:
     # -- create path ----------------------------------------------------------
 
     def _prepare_allocation(self, cr: dict, alloc: dict) -> Optional[dict]:
-        """Realize one `creating` allocation; returns the prepared-entry dict
-        keyed by partition uuid, or None if it must be retried later."""
+        """Realize one `creating` allocation on the DEVICE (mode set + ordinal
+        resolution); returns the prepared-entry dict keyed by partition uuid,
+        or None if it must be retried later. Store writes happen in one
+        batched round-trip in the caller (_commit_prepare)."""
         gpu_uuid = alloc["gpuUUID"]
         self._ensure_gpu_mode(cr, gpu_uuid, alloc["computeMode"], alloc["memoryMode"])
         g = self._gpus[gpu_uuid]
@@ -240,32 +248,6 @@ class NodeAgent:
             raise SmiError(
                 f"gpu {gpu_uuid}: ordinal {ordinal} not present in mode {g.compute_mode}"
             )
-        # the workload contract: a pod-named ConfigMap delivering the visible
-        # partition (reference: createConfigMap with NVIDIA_/CUDA_VISIBLE_DEVICES,
-        # instaslice_daemonset.go:796-818 -> ROCR_/HIP_ here)
-        # ROCR accepts device ordinals (or rocminfo-style GPU-<id> strings),
-        # NOT amdsmi UUIDs — verified on MI355X: a UUID value yields "no
-        # ROCm-capable device is detected". The partition's node-wide HIP
-        # ordinal is the selector; the amdsmi UUID rides along for operators.
-        cm = {
-            "apiVersion": "v1",
-            "kind": "ConfigMap",
-            "metadata": {"name": alloc["podName"], "namespace": alloc["namespace"]},
-            "data": {
-                "ROCR_VISIBLE_DEVICES": str(part.device_index),
-                "HIP_VISIBLE_DEVICES": str(part.device_index),
-                "INSTASLICE_PARTITION_UUID": part.uuid,
-                "INSTASLICE_PARTITION_ORDINAL": str(ordinal),
-                "INSTASLICE_PARTITION_GPU": gpu_uuid,
-            },
-        }
-        try:
-            self.store.create(cm)
-        except AlreadyExists:
-            pass
-        # node capacity pinning (reference: createInstaSliceResource,
-        # instaslice_daemonset.go:277-300)
-        self._patch_capacity(alloc["podName"], add=True)
         prep = PreparedDetails(
             parent_gpu_uuid=gpu_uuid,
             ordinal=ordinal,
@@ -278,21 +260,69 @@ class NodeAgent:
         )
         return {part.uuid: prep.to_dict()}
 
-    def _patch_capacity(self, pod_name: str, add: bool) -> None:
-        def mut(node: dict) -> Optional[dict]:
-            cap = node.setdefault("status", {}).setdefault("capacity", {})
-            key = POD_RESOURCE_PREFIX + pod_name
-            if add:
-                if cap.get(key) == 1:
-                    return None
-                cap[key] = 1
-            else:
-                if key not in cap:
-                    return None
-                del cap[key]
-            return node
-
-        self.store.update_with_retry("Node", self.node_name, "", mut)
+    def _commit_prepare(self, alloc: dict, prepared_entry: dict) -> bool:
+        """ONE batched store round-trip for the whole create-path commit:
+        pod ConfigMap (visible-devices contract, reference createConfigMap
+        instaslice_daemonset.go:796-818 -> ROCR_/HIP_ here) + node capacity
+        pin (createInstaSliceResource, :277-300) + CR status flip creating ->
+        created with the Prepared entry. Returns False if the allocation
+        changed under us (raced teardown)."""
+        part_uuid, prep = next(iter(prepared_entry.items()))
+        gpu_uuid = alloc["gpuUUID"]
+        g = self._gpus[gpu_uuid]
+        # ROCR accepts device ordinals (or rocminfo-style GPU-<id> strings),
+        # NOT amdsmi UUIDs — verified on MI355X: a UUID value yields "no
+        # ROCm-capable device is detected". The partition's node-wide HIP
+        # ordinal is the selector; the amdsmi UUID rides along for operators.
+        cm = {
+            "apiVersion": "v1",
+            "kind": "ConfigMap",
+            "metadata": {"name": alloc["podName"], "namespace": alloc["namespace"]},
+            "data": {
+                "ROCR_VISIBLE_DEVICES": str(prep["deviceIndex"]),
+                "HIP_VISIBLE_DEVICES": str(prep["deviceIndex"]),
+                "INSTASLICE_PARTITION_UUID": part_uuid,
+                "INSTASLICE_PARTITION_ORDINAL": str(alloc["ordinal"]),
+                "INSTASLICE_PARTITION_GPU": gpu_uuid,
+            },
+        }
+        pu = alloc["podUUID"]
+        res = self.store.batch([
+            {"verb": "create", "obj": cm},
+            {"verb": "patch", "kind": "Node", "name": self.node_name,
+             "namespace": "", "ops": [
+                 {"op": "set",
+                  "path": ["status", "capacity", POD_RESOURCE_PREFIX + alloc["podName"]],
+                  "value": 1},
+             ]},
+            {"verb": "patch", "kind": "Instaslice", "name": self.node_name,
+             "namespace": INSTASLICE_NS, "ops": [
+                 {"op": "test",
+                  "path": ["spec", "allocations", pu, "allocationStatus"],
+                  "value": AllocationStatus.CREATING},
+                 {"op": "set",
+                  "path": ["spec", "allocations", pu, "allocationStatus"],
+                  "value": AllocationStatus.CREATED},
+                 {"op": "merge", "path": ["spec", "prepared"],
+                  "value": prepared_entry},
+                 {"op": "set", "path": ["spec", "gpus", gpu_uuid, "computeMode"],
+                  "value": g.compute_mode},
+                 {"op": "set", "path": ["spec", "gpus", gpu_uuid, "memoryMode"],
+                  "value": g.memory_mode},
+                 {"op": "add_to_set",
+                  "path": ["spec", "gpus", gpu_uuid, "usedOrdinals"],
+                  "value": alloc["ordinal"]},
+             ]},
+        ], quiet=True)
+        if not res[0]["ok"] and res[0]["error"]["type"] != "AlreadyExists":
+            self.log.warning("configmap create failed: %s", res[0]["error"])
+        commit = res[2]
+        if not commit["ok"]:
+            # Conflict: status moved (pod deleted mid-create) — the event for
+            # that change re-reconciles us with fresh state
+            self.log.debug("prepare commit superseded: %s", commit["error"]["msg"])
+            return False
+        return True
 
     def _needs_mode_change(self, alloc: dict) -> bool:
         g = self._gpus.get(alloc["gpuUUID"])
@@ -301,41 +331,40 @@ class NodeAgent:
     def _fail_allocation(self, pod_uuid: str, gpu_uuid: str, lock_mode: bool) -> None:
         """Flip the allocation to `failed` (controller will re-place) and
         optionally mark the GPU mode-locked in the CR."""
-
-        def mut(obj: dict) -> Optional[dict]:
-            spec = obj["spec"]
-            alloc = (spec.get("allocations") or {}).get(pod_uuid)
-            changed = False
-            if alloc and alloc["allocationStatus"] == AllocationStatus.CREATING:
-                alloc["allocationStatus"] = AllocationStatus.FAILED
-                changed = True
-            gd = (spec.get("gpus") or {}).get(gpu_uuid)
-            if lock_mode and gd and not gd.get("modeLocked"):
-                gd["modeLocked"] = True
-                changed = True
-            return obj if changed else None
-
-        self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, mut)
+        reqs = [
+            {"verb": "patch", "kind": "Instaslice", "name": self.node_name,
+             "namespace": INSTASLICE_NS, "ops": [
+                 {"op": "test",
+                  "path": ["spec", "allocations", pod_uuid, "allocationStatus"],
+                  "value": AllocationStatus.CREATING},
+                 {"op": "set",
+                  "path": ["spec", "allocations", pod_uuid, "allocationStatus"],
+                  "value": AllocationStatus.FAILED},
+             ]},
+        ]
+        if lock_mode:
+            reqs.append(
+                {"verb": "patch", "kind": "Instaslice", "name": self.node_name,
+                 "namespace": INSTASLICE_NS, "ops": [
+                     {"op": "set", "path": ["spec", "gpus", gpu_uuid, "modeLocked"],
+                      "value": True},
+                 ]})
+        self.store.batch(reqs, quiet=True)
         from instaslice_amd.metrics import get_metrics
 
         get_metrics().allocation("failed")
 
-    # -- delete path ----------------------------------------------------------
-
-    def _teardown_allocation(self, cr: dict, pod_uuid: str, alloc: dict) -> None:
-        try:
-            self.store.delete("ConfigMap", alloc["podName"], alloc["namespace"])
-        except NotFound:
-            pass
-        self._patch_capacity(alloc["podName"], add=False)
-
     # -- reconcile ------------------------------------------------------------
 
     def _reconcile(self, key: Key) -> Result:
-        try:
-            cr = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
-        except NotFound:
-            return Result()
+        # informer cache first (fed by our filtered watch): on the TCP store
+        # this removes one GET round-trip from every reconcile
+        cr = self.engine.cached(("Instaslice", INSTASLICE_NS, self.node_name))
+        if cr is None:
+            try:
+                cr = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+            except NotFound:
+                return Result()
         allocations = cr.get("spec", {}).get("allocations") or {}
         requeue: Optional[float] = None
 
@@ -384,57 +413,58 @@ class NodeAgent:
                     requeue = REQUEUE_BUSY_S
                     continue
                 part_uuid = next(iter(prepared_entry))
+                if self._commit_prepare(alloc, prepared_entry):
+                    from instaslice_amd.metrics import get_metrics
 
-                def commit(obj: dict, pu=pod_uuid, pe=prepared_entry, a=alloc) -> Optional[dict]:
-                    spec = obj["spec"]
-                    cur = (spec.get("allocations") or {}).get(pu)
-                    if not cur or cur["allocationStatus"] != AllocationStatus.CREATING:
-                        return None
-                    cur["allocationStatus"] = AllocationStatus.CREATED
-                    spec.setdefault("prepared", {}).update(pe)
-                    gd = spec["gpus"][a["gpuUUID"]]
-                    g = self._gpus[a["gpuUUID"]]
-                    used = set(gd.get("usedOrdinals", []))
-                    used.add(a["ordinal"])
-                    gd.update(
-                        computeMode=g.compute_mode,
-                        memoryMode=g.memory_mode,
-                        usedOrdinals=sorted(used),
+                    get_metrics().allocation("created")
+                    self.log.debug(
+                        "prepared partition %s for pod %s",
+                        part_uuid[:8], alloc["podName"],
                     )
-                    return obj
-
-                self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, commit)
-                from instaslice_amd.metrics import get_metrics
-
-                get_metrics().allocation("created")
-                self.log.debug(
-                    "prepared partition %s for pod %s", part_uuid[:8], alloc["podName"]
-                )
             elif status == AllocationStatus.DELETED:
                 self._teardown_allocation(cr, pod_uuid, alloc)
-
-                def cleanup(obj: dict, pu=pod_uuid, a=alloc) -> Optional[dict]:
-                    spec = obj["spec"]
-                    if pu not in (spec.get("allocations") or {}):
-                        return None
-                    del spec["allocations"][pu]
-                    prepared = spec.get("prepared") or {}
-                    for puid in [k for k, v in prepared.items() if v["podUUID"] == pu]:
-                        del prepared[puid]
-                    gd = spec["gpus"].get(a["gpuUUID"])
-                    if gd:
-                        used = set(gd.get("usedOrdinals", []))
-                        used.discard(a["ordinal"])
-                        gd["usedOrdinals"] = sorted(used)
-                    return obj
-
-                self.store.update_with_retry("Instaslice", self.node_name, INSTASLICE_NS, cleanup)
-                from instaslice_amd.metrics import get_metrics
-
-                get_metrics().allocation("deleted")
                 if self.reset_mode_on_empty:
                     self._maybe_reset_gpu(alloc["gpuUUID"])
         return Result(requeue_after=requeue)
+
+    # -- delete path ----------------------------------------------------------
+
+    def _teardown_allocation(self, cr: dict, pod_uuid: str, alloc: dict) -> None:
+        """ONE batched round-trip for the whole delete path: drop ConfigMap +
+        capacity pin + allocation/prepared/ordinal from the CR (reference does
+        these as separate API calls, instaslice_daemonset.go:415-470)."""
+        part_uuids = [
+            k for k, v in (cr["spec"].get("prepared") or {}).items()
+            if v["podUUID"] == pod_uuid
+        ]
+        cr_ops = [
+            {"op": "test",
+             "path": ["spec", "allocations", pod_uuid, "allocationStatus"],
+             "value": AllocationStatus.DELETED},
+            {"op": "delete", "path": ["spec", "allocations", pod_uuid]},
+            {"op": "remove_from_set",
+             "path": ["spec", "gpus", alloc["gpuUUID"], "usedOrdinals"],
+             "value": alloc["ordinal"]},
+        ] + [
+            {"op": "delete", "path": ["spec", "prepared", puid]}
+            for puid in part_uuids
+        ]
+        res = self.store.batch([
+            {"verb": "delete", "kind": "ConfigMap", "name": alloc["podName"],
+             "namespace": alloc["namespace"]},
+            {"verb": "patch", "kind": "Node", "name": self.node_name,
+             "namespace": "", "ops": [
+                 {"op": "delete",
+                  "path": ["status", "capacity",
+                           POD_RESOURCE_PREFIX + alloc["podName"]]},
+             ]},
+            {"verb": "patch", "kind": "Instaslice", "name": self.node_name,
+             "namespace": INSTASLICE_NS, "ops": cr_ops},
+        ], quiet=True)
+        if res[2]["ok"]:
+            from instaslice_amd.metrics import get_metrics
+
+            get_metrics().allocation("deleted")
 
     def _maybe_reset_gpu(self, gpu_uuid: str) -> None:
         """Reference-parity teardown (ci/gi Destroy analog,
@@ -471,13 +501,11 @@ class NodeAgent:
 
     def _heartbeat_loop(self) -> None:
         while not self._hb_stop.is_set():
-            def beat(obj: dict) -> Optional[dict]:
-                obj.setdefault("status", {})["heartbeat"] = time.time()
-                return obj
-
             try:
-                self.store.update_with_retry(
-                    "Instaslice", self.node_name, INSTASLICE_NS, beat
+                self.store.patch(
+                    "Instaslice", self.node_name, INSTASLICE_NS,
+                    [{"op": "set", "path": ["status", "heartbeat"],
+                      "value": time.time()}], quiet=True,
                 )
             except Exception as e:  # store outage: keep trying
                 self.log.warning("heartbeat failed: %s", e)

@@ -215,6 +215,7 @@ def new_pod(
     node_selector: Optional[Dict[str, str]] = None,
     gated: bool = True,
     group: Optional[str] = None,
+    labels: Optional[Dict[str, str]] = None,
 ) -> dict:
     """Synthetic gated pod following the reference's consumer contract
     (samples/test-pod.yaml:1-21): scheduling gate + finalizer + profile limit
@@ -238,6 +239,7 @@ def new_pod(
             "annotations": (
                 {"org.instaslice/group": group} if group else {}
             ),
+            "labels": labels or {},
         },
         "spec": {
             "schedulingGates": [{"name": GATE_NAME}] if gated else [],

@@ -241,25 +241,50 @@ class PodController:
             "Pod", md["name"], md.get("namespace", "default"), mut
         )
 
-    def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str) -> None:
-        def mut(cr: dict) -> Optional[dict]:
-            alloc = (cr.get("spec", {}).get("allocations") or {}).get(pod_uid)
-            if not alloc or alloc["allocationStatus"] == status:
-                return None
-            alloc["allocationStatus"] = status
-            return cr
+    def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str,
+                               expect: Optional[str] = None) -> None:
+        from instaslice_amd.store.memstore import Conflict
 
-        self.store.update_with_retry("Instaslice", cr_name, INSTASLICE_NS, mut)
+        ops = []
+        if expect is not None:
+            ops.append({"op": "test",
+                        "path": ["spec", "allocations", pod_uid, "allocationStatus"],
+                        "value": expect})
+        else:
+            # require the allocation to exist (its podUUID key is itself)
+            ops.append({"op": "test",
+                        "path": ["spec", "allocations", pod_uid, "podUUID"],
+                        "value": pod_uid})
+        ops.append({"op": "set",
+                    "path": ["spec", "allocations", pod_uid, "allocationStatus"],
+                    "value": status})
+        try:
+            self.store.patch("Instaslice", cr_name, INSTASLICE_NS, ops)
+        except (Conflict, NotFound):
+            pass  # allocation gone or moved on; events re-reconcile us
 
     # -- reconcile ------------------------------------------------------------
 
+    def _crs(self) -> List[dict]:
+        """Current Instaslice CRs from the informer cache (kind-wide watch
+        keeps it complete after replay; zero store round-trips, zero
+        per-reconcile snapshot cost). Placement decisions taken on a cached
+        CR are re-validated against the FRESH object inside add_alloc's
+        update cycle, so staleness cannot double-book an ordinal."""
+        crs = self.engine.cached_list("Instaslice")
+        if crs:
+            return crs
+        return self.store.list("Instaslice")  # pre-replay startup window
+
     def _reconcile(self, key: Key) -> Result:
         _, namespace, name = key
-        try:
-            pod = self.store.get("Pod", name, namespace)
-        except NotFound:
-            self._unschedulable_keys.discard((namespace, name))
-            return Result()
+        pod = self.engine.cached(("Pod", namespace, name))
+        if pod is None:
+            try:
+                pod = self.store.get("Pod", name, namespace)
+            except NotFound:
+                self._unschedulable_keys.discard((namespace, name))
+                return Result()
         md = pod["metadata"]
         uid = md["uid"]
 
@@ -271,7 +296,7 @@ class PodController:
                 elapsed = time.time() - float(md["deletionTimestamp"])
                 if elapsed < self.teardown_grace_s:
                     return Result(requeue_after=self.teardown_grace_s - elapsed)
-            found = self._find_allocation(uid, self.store.list("Instaslice"))
+            found = self._find_allocation(uid, self._crs())
             if found:
                 cr, alloc = found
                 if alloc["allocationStatus"] != AllocationStatus.DELETED:
@@ -293,7 +318,7 @@ class PodController:
             return Result()
 
         # admission path
-        crs = self.store.list("Instaslice")
+        crs = self._crs()
         found = self._find_allocation(uid, crs)
         if found is None:
             limits = pod_limits(pod)
@@ -327,22 +352,52 @@ class PodController:
             )
             return Result(requeue_after=0.01)
         if status == AllocationStatus.CREATED:
-            # agent realized the partition: let the pod schedule
-            def do_ungate(p: dict) -> Optional[dict]:
-                new = ungate_pod(p)
-                if new["spec"]["schedulingGates"] == p["spec"].get("schedulingGates"):
-                    return None
-                # gate removal unblocks kube-scheduler; reflect it in status
-                new["status"]["phase"] = "Pending"
-                new["status"]["conditions"] = [
-                    {"type": "PodScheduled", "status": "True", "message": "ungated"}
-                ]
-                return new
+            # agent realized the partition: let the pod schedule. One batched
+            # round-trip: ungate the pod AND flip the allocation to ungated.
+            from instaslice_amd import GATE_NAME
 
-            self.store.update_with_retry("Pod", name, namespace, do_ungate)
-            self._set_allocation_status(
-                cr["metadata"]["name"], uid, AllocationStatus.UNGATED
-            )
+            res = self.store.batch([
+                {"verb": "patch", "kind": "Pod", "name": name,
+                 "namespace": namespace, "ops": [
+                     {"op": "test", "path": ["spec", "schedulingGates"],
+                      "value": [{"name": GATE_NAME}]},
+                     {"op": "set", "path": ["spec", "schedulingGates"],
+                      "value": []},
+                     # gate removal unblocks kube-scheduler; reflect in status
+                     {"op": "set", "path": ["status", "phase"],
+                      "value": "Pending"},
+                     {"op": "set", "path": ["status", "conditions"],
+                      "value": [{"type": "PodScheduled", "status": "True",
+                                 "message": "ungated"}]},
+                 ]},
+                {"verb": "patch", "kind": "Instaslice",
+                 "name": cr["metadata"]["name"], "namespace": INSTASLICE_NS,
+                 "ops": [
+                     {"op": "test",
+                      "path": ["spec", "allocations", uid, "allocationStatus"],
+                      "value": AllocationStatus.CREATED},
+                     {"op": "set",
+                      "path": ["spec", "allocations", uid, "allocationStatus"],
+                      "value": AllocationStatus.UNGATED},
+                 ]},
+            ])
+            if not res[0]["ok"] and res[0]["error"]["type"] == "Conflict":
+                # gate list isn't exactly ours (extra gates / already ungated):
+                # fall back to the precise read-modify-write
+                def do_ungate(p: dict) -> Optional[dict]:
+                    new = ungate_pod(p)
+                    if new["spec"]["schedulingGates"] == p["spec"].get(
+                        "schedulingGates"
+                    ):
+                        return None
+                    new["status"]["phase"] = "Pending"
+                    new["status"]["conditions"] = [
+                        {"type": "PodScheduled", "status": "True",
+                         "message": "ungated"}
+                    ]
+                    return new
+
+                self.store.update_with_retry("Pod", name, namespace, do_ungate)
             t0 = self._pending_since.pop(uid, None)
             if t0 is not None:
                 dt = time.monotonic() - t0

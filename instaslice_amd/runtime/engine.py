@@ -49,6 +49,10 @@ class WatchSpec:
     # Reference analog: podMapFunc mapping Instaslice events to pods
     # (instaslice_controller.go:398-407).
     map_fn: Optional[Callable[[str, dict], List[Key]]] = None
+    # server-side scoping (store watch filters): a node agent subscribes to
+    # its OWN Instaslice CR only, so store event fan-out stays O(relevant)
+    # instead of O(cluster) — the k8s field-selector analog
+    filters: Optional[List[dict]] = None
 
 
 class Engine:
@@ -73,6 +77,14 @@ class Engine:
 
         self._queue: List[Key] = []
         self._queued: set = set()
+        # informer cache: latest object seen per key, fed by the watch pumps
+        # (client-go shared-informer analog). Reconciles read it via
+        # `cached()` instead of a store GET — on the TCP store that removes
+        # one round-trip from every reconcile. Writers guard their patches
+        # with test ops, so a stale cache surfaces as Conflict, after which
+        # `invalidate()` forces the next reconcile back to a real GET.
+        self._cache: Dict[Key, dict] = {}
+        self._cache_lock = threading.Lock()
         # keys currently being reconciled (multi-worker safety: a key is
         # never processed concurrently; events arriving mid-reconcile mark it
         # dirty and it re-runs right after — controller-runtime semantics)
@@ -134,12 +146,36 @@ class Engine:
                 continue
             event_type, obj = ev
             md = obj.get("metadata", {})
+            own_key = (obj["kind"], md.get("namespace", ""), md["name"])
+            with self._cache_lock:
+                if event_type == "DELETED":
+                    self._cache.pop(own_key, None)
+                else:
+                    self._cache[own_key] = obj
             if spec.map_fn is None:
-                keys = [(obj["kind"], md.get("namespace", ""), md["name"])]
+                keys = [own_key]
             else:
                 keys = spec.map_fn(event_type, obj)
             for k in keys:
                 self.enqueue(k)
+
+    def cached(self, key: Key) -> Optional[dict]:
+        """Latest watched object for `key`, or None. READ-ONLY — watch event
+        objects are shared; mutate a copy."""
+        with self._cache_lock:
+            return self._cache.get(key)
+
+    def cached_list(self, kind: str) -> List[dict]:
+        """All cached objects of `kind`, name-sorted. READ-ONLY objects.
+        Complete once the watch replay has been pumped; callers that can see
+        an empty startup window must tolerate it (reconcile requeues cover
+        it, same as a not-yet-created CR)."""
+        with self._cache_lock:
+            return [o for k, o in sorted(self._cache.items()) if k[0] == kind]
+
+    def invalidate(self, key: Key) -> None:
+        with self._cache_lock:
+            self._cache.pop(key, None)
 
     def _work(self) -> None:
         from instaslice_amd.metrics import get_metrics
@@ -187,7 +223,7 @@ class Engine:
 
     def start(self) -> "Engine":
         for spec in self.watch_specs:
-            w = self.store.watch(spec.kind, replay=True)
+            w = self.store.watch(spec.kind, replay=True, filters=spec.filters)
             self._watches.append(w)
             t = threading.Thread(
                 target=self._pump, args=(spec, w), name=f"{self.name}-pump-{spec.kind}",

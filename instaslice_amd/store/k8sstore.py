@@ -211,9 +211,38 @@ class K8sStore:
         "Node": ("core", "list_node"),
     }
 
-    def watch(self, kind: Optional[str] = None, *, replay: bool = True):
-        if kind is None:
+    def watch(self, kind: Optional[str] = None, *, replay: bool = True,
+              filters: Optional[list] = None):
+        if kind is None and not filters:
             raise ValueError("K8sStore.watch requires an explicit kind")
+        if filters:
+            # one upstream watch per distinct kind named in the filters,
+            # client-side filtered (k8s field selectors could narrow further;
+            # correctness first)
+            from instaslice_amd.store.memstore import _filter_matches
+
+            kinds = {f.get("kind") for f in filters}
+            if None in kinds:
+                raise ValueError("K8sStore filters need explicit kinds")
+            merged = _K8sWatch()
+            merged.kind = None
+            for knd in sorted(kinds):
+                inner = self.watch(knd, replay=replay)
+
+                def forward(inner=inner):
+                    while not merged._stopped:
+                        ev = inner.next(timeout=0.5)
+                        if ev is None:
+                            continue
+                        if any(_filter_matches(f, ev[1]) for f in filters):
+                            merged._q.put(ev)
+                    inner.stop()
+
+                t = threading.Thread(target=forward, daemon=True,
+                                     name=f"k8swatch-filter-{knd}")
+                t.start()
+                self._watch_threads.append(t)
+            return merged
         w = _K8sWatch()
         w.kind = kind
 
@@ -245,6 +274,47 @@ class K8sStore:
         t.start()
         self._watch_threads.append(t)
         return w
+
+    def patch(self, kind: str, name: str, namespace: str = "",
+              ops: Optional[list] = None, *, quiet: bool = False) -> dict:
+        """PATCH emulation over get+update (real k8s rejects unknown-verb
+        shortcuts; server-side-apply would be the native path). Failed test
+        ops raise Conflict straight through — same contract as MemStore."""
+        from instaslice_amd.store.memstore import apply_patch_ops
+
+        def mut(obj: dict):
+            apply_patch_ops(obj, ops or [])
+            return obj
+
+        res = self.update_with_retry(kind, name, namespace, mut)
+        if res is None:
+            raise NotFound(f"{kind}/{name} not found")
+        return res
+
+    def batch(self, requests: list, *, quiet: bool = False) -> list:
+        out = []
+        for req in requests:
+            verb = req.get("verb")
+            try:
+                if verb == "create":
+                    res = self.create(req["obj"])
+                elif verb == "get":
+                    res = self.get(req["kind"], req["name"], req.get("namespace", ""))
+                elif verb == "update":
+                    res = self.update(req["obj"])
+                elif verb == "delete":
+                    self.delete(req["kind"], req["name"], req.get("namespace", ""))
+                    res = None
+                elif verb == "patch":
+                    res = self.patch(req["kind"], req["name"],
+                                     req.get("namespace", ""), req.get("ops"))
+                else:
+                    raise ValueError(f"unknown batch verb {verb!r}")
+                out.append({"ok": True, "result": res})
+            except (Conflict, NotFound, AlreadyExists) as e:
+                out.append({"ok": False,
+                            "error": {"type": type(e).__name__, "msg": str(e)}})
+        return out
 
     def update_with_retry(
         self, kind: str, name: str, namespace: str,

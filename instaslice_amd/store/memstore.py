@@ -55,17 +55,112 @@ def _key(obj: dict) -> Key:
     return (obj["kind"], md.get("namespace", ""), md["name"])
 
 
+def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
+    """Apply patch ops to `obj` in place (shared by MemStore.patch and the
+    K8sStore emulation). Ops, applied in order:
+
+      {"op":"set",    "path":[...], "value":v}   set nested key (mkdir -p)
+      {"op":"merge",  "path":[...], "value":{}}  dict.update at path
+      {"op":"delete", "path":[...]}              remove key if present
+      {"op":"add_to_set",      "path":[...], "value":v}  sorted-list add
+      {"op":"remove_from_set", "path":[...], "value":v}  sorted-list drop
+      {"op":"test",   "path":[...], "value":v}   require equality
+      {"op":"test",   "path":[...], "absent":true}  require key missing
+
+    Raises Conflict on a failed test. Callers provide all-or-nothing by
+    applying to a copy and committing only on success."""
+    for op in ops:
+        path = op["path"]
+        kind_op = op["op"]
+        if kind_op == "test":
+            node, missing = obj, False
+            for p in path:
+                if not isinstance(node, dict) or p not in node:
+                    missing = True
+                    break
+                node = node[p]
+            if op.get("absent"):
+                if not missing:
+                    raise Conflict(f"patch test: {path} expected absent")
+            elif missing or node != op.get("value"):
+                raise Conflict(f"patch test: {path} != {op.get('value')!r}")
+            continue
+        # navigate to parent, creating dicts along the way
+        node = obj
+        for p in path[:-1]:
+            nxt = node.get(p) if isinstance(node, dict) else None
+            if not isinstance(nxt, (dict, list)):
+                nxt = {}
+                node[p] = nxt
+            node = nxt
+        leaf = path[-1]
+        if kind_op == "set":
+            node[leaf] = op["value"]
+        elif kind_op == "merge":
+            tgt = node.get(leaf)
+            if not isinstance(tgt, dict):
+                tgt = {}
+                node[leaf] = tgt
+            tgt.update(op["value"])
+        elif kind_op == "delete":
+            if isinstance(node, dict):
+                node.pop(leaf, None)
+        elif kind_op == "add_to_set":
+            cur = node.get(leaf)
+            if not isinstance(cur, list):
+                cur = []
+            if op["value"] not in cur:
+                cur = sorted(cur + [op["value"]])
+            node[leaf] = cur
+        elif kind_op == "remove_from_set":
+            cur = node.get(leaf)
+            if isinstance(cur, list) and op["value"] in cur:
+                node[leaf] = [x for x in cur if x != op["value"]]
+        else:
+            raise ValueError(f"unknown patch op {kind_op!r}")
+    return obj
+
+
+def _filter_matches(f: dict, obj: dict) -> bool:
+    if f.get("kind") is not None and f["kind"] != obj["kind"]:
+        return False
+    md = obj.get("metadata", {})
+    if f.get("name") is not None and f["name"] != md.get("name"):
+        return False
+    if f.get("namespace") is not None and f["namespace"] != md.get("namespace", ""):
+        return False
+    labels = f.get("labels")
+    if labels:
+        have = md.get("labels") or {}
+        for k, v in labels.items():
+            if have.get(k) != v:
+                return False
+    return True
+
+
 class Watch:
     """A subscription delivering (event_type, object) tuples.
 
     event_type is "ADDED" | "MODIFIED" | "DELETED". Objects are deep copies.
+    `filters` (list of {kind,name,namespace,labels} dicts, OR-combined) scope
+    the subscription server-side — the k8s field/label-selector analog. A
+    cluster-wide all-kinds watch is O(total event rate); filtered watches keep
+    per-client traffic O(own events), which is what lets N node agents share
+    one store without N^2 event fan-out.
     """
 
-    def __init__(self, store: "MemStore", kind: Optional[str]):
+    def __init__(self, store: "MemStore", kind: Optional[str],
+                 filters: Optional[List[dict]] = None):
         self._store = store
         self.kind = kind
+        self.filters = filters
         self._q: "queue.Queue[Optional[Tuple[str, dict]]]" = queue.Queue()
         self._stopped = False
+
+    def _matches(self, obj: dict) -> bool:
+        if self.filters is not None:
+            return any(_filter_matches(f, obj) for f in self.filters)
+        return self.kind is None or self.kind == obj["kind"]
 
     def _push(self, event: Tuple[str, dict]) -> None:
         if not self._stopped:
@@ -258,16 +353,77 @@ class MemStore:
                 self._notify("DELETED", obj)
             self._mark_dirty()
 
+    # -- patch ------------------------------------------------------------
+
+    def patch(self, kind: str, name: str, namespace: str = "",
+              ops: Optional[List[dict]] = None, *, quiet: bool = False):
+        """Atomic server-side partial update — the k8s PATCH analog, and the
+        latency-critical verb: one round-trip replaces the get-mutate-update
+        cycle (and its Conflict retries) for the hot reconcile paths.
+        Op grammar in `apply_patch_ops`. A failed test raises Conflict (no
+        mutation happens) — callers treat it as "state moved on; the
+        event-driven reconcile will re-run"."""
+        with self._lock:
+            k = (kind, namespace, name)
+            if k not in self._objects:
+                raise NotFound(f"{k} not found")
+            obj = _snapshot(self._objects[k])
+            apply_patch_ops(obj, ops or [])
+            self._rv += 1
+            obj["metadata"]["resourceVersion"] = str(self._rv)
+            self._objects[k] = obj
+            if obj["metadata"].get("deletionTimestamp") and not obj["metadata"].get(
+                "finalizers"
+            ):
+                del self._objects[k]
+                self._notify("DELETED", obj)
+            else:
+                self._notify("MODIFIED", obj)
+            self._mark_dirty()
+            return None if quiet else _snapshot(obj)
+
+    def batch(self, requests: List[dict], *, quiet: bool = False) -> List[dict]:
+        """Execute several verbs in one call; per-entry results so callers
+        keep the same error semantics as individual calls. Over the TCP store
+        this collapses the agent's serial create-path writes (ConfigMap +
+        Node capacity + CR commit) into a single round-trip."""
+        out: List[dict] = []
+        for req in requests:
+            verb = req.get("verb")
+            try:
+                if verb == "create":
+                    res = self.create(req["obj"])
+                elif verb == "get":
+                    res = self.get(req["kind"], req["name"], req.get("namespace", ""))
+                elif verb == "update":
+                    res = self.update(req["obj"])
+                elif verb == "delete":
+                    self.delete(req["kind"], req["name"], req.get("namespace", ""))
+                    res = None
+                elif verb == "patch":
+                    res = self.patch(req["kind"], req["name"],
+                                     req.get("namespace", ""), req.get("ops"))
+                else:
+                    raise ValueError(f"unknown batch verb {verb!r}")
+                out.append({"ok": True, "result": None if quiet else res})
+            except (Conflict, NotFound, AlreadyExists) as e:
+                out.append({"ok": False, "result": None,
+                            "error": {"type": type(e).__name__, "msg": str(e)}})
+        return out
+
     # -- watch ------------------------------------------------------------
 
-    def watch(self, kind: Optional[str] = None, *, replay: bool = True) -> Watch:
-        """Subscribe to events for `kind` (None = all kinds). With replay,
-        current objects are delivered first as ADDED (k8s informer analog)."""
+    def watch(self, kind: Optional[str] = None, *, replay: bool = True,
+              filters: Optional[List[dict]] = None) -> Watch:
+        """Subscribe to events for `kind` (None = all kinds), or — with
+        `filters` — to the union of {kind,name,namespace,labels} selectors.
+        With replay, current objects are delivered first as ADDED (k8s
+        informer analog)."""
         with self._lock:
-            w = Watch(self, kind)
+            w = Watch(self, kind, filters)
             if replay:
-                for (knd, _, _), o in sorted(self._objects.items()):
-                    if kind is None or knd == kind:
+                for _, o in sorted(self._objects.items()):
+                    if w._matches(o):
                         w._push(("ADDED", _snapshot(o)))
             self._watches.append(w)
             return w
@@ -280,7 +436,7 @@ class MemStore:
             if w._stopped:
                 self._watches.remove(w)
                 continue
-            if w.kind is None or w.kind == obj["kind"]:
+            if w._matches(obj):
                 if snap is None:
                     snap = _snapshot(obj)
                 w._push((event_type, snap))

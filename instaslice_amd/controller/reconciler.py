@@ -57,9 +57,21 @@ class PodController:
         requeue_no_fit_s: float = REQUEUE_NO_FIT_S,
         node_stale_after_s: float = 30.0,
         workers: int = 2,
+        shard_index: int = 0,
+        shard_count: int = 1,
     ) -> None:
         self.store = store
         self.policy: AllocationPolicy = get_policy(policy)
+        # horizontal sharding: controller instance i of k owns the pods whose
+        # crc32(ns/name) % k == i. Python reconciles are GIL-bound, so the
+        # scale-out axis is processes, not worker threads; shards race only
+        # on CR writes, where add_alloc's fresh-object re-validation +
+        # resourceVersion conflicts already arbitrate (same mechanism as
+        # concurrent workers). The reference runs exactly one controller
+        # (leader election, cmd/controller/main.go:107-108) — sharding is a
+        # deliberate scale-out improvement over that.
+        self.shard_index = shard_index
+        self.shard_count = max(1, shard_count)
         self.teardown_grace_s = teardown_grace_s
         self.requeue_no_fit_s = requeue_no_fit_s
         # failure detection: nodes whose agent heartbeat is older than this
@@ -79,15 +91,32 @@ class PodController:
         # placement races between different pods are caught by the
         # re-validation inside add_alloc + resourceVersion conflicts
         self.engine = Engine(
-            name="controller",
+            name=f"controller-{shard_index}" if shard_count > 1 else "controller",
             store=store,
             reconcile=self._reconcile,
             watches=[
-                WatchSpec(kind="Pod"),
+                WatchSpec(kind="Pod", map_fn=self._own_pods_only),
                 WatchSpec(kind="Instaslice", map_fn=self._instaslice_to_pods),
             ],
             workers=workers,
         )
+
+    # -- sharding ----------------------------------------------------------
+
+    def _owns(self, namespace: str, name: str) -> bool:
+        if self.shard_count <= 1:
+            return True
+        import zlib
+
+        h = zlib.crc32(f"{namespace}/{name}".encode())
+        return h % self.shard_count == self.shard_index
+
+    def _own_pods_only(self, event_type: str, obj: dict) -> List[Key]:
+        md = obj["metadata"]
+        ns, name = md.get("namespace", ""), md["name"]
+        if not self._owns(ns, name):
+            return []
+        return [("Pod", ns, name)]
 
     # -- watch mapping ------------------------------------------------------
     # Reference: podMapFunc maps an Instaslice with "created" allocations back
@@ -99,7 +128,7 @@ class PodController:
             if alloc.get("allocationStatus") in (
                 AllocationStatus.CREATED,
                 AllocationStatus.FAILED,
-            ):
+            ) and self._owns(alloc["namespace"], alloc["podName"]):
                 keys.append(("Pod", alloc["namespace"], alloc["podName"]))
         # freed/changed capacity: give waiting pods another placement pass
         keys.extend(("Pod", ns, name) for (ns, name) in list(self._unschedulable_keys))

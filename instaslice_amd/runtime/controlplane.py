@@ -17,6 +17,41 @@ import sys
 from typing import Optional
 
 
+def _run_controller_shard(
+    port: int,
+    policy: str,
+    teardown_grace_s: float,
+    requeue_no_fit_s: float,
+    workers: int,
+    shard_index: int,
+    shard_count: int,
+    conn,
+) -> None:
+    """multiprocessing target: one controller shard over TCP."""
+    sys.setswitchinterval(0.001)
+    from instaslice_amd.controller.reconciler import PodController
+    from instaslice_amd.store.netstore import NetStoreClient
+
+    store = NetStoreClient("127.0.0.1", port)
+    controller = PodController(
+        store,
+        policy=policy,
+        teardown_grace_s=teardown_grace_s,
+        requeue_no_fit_s=requeue_no_fit_s,
+        workers=workers,
+        shard_index=shard_index,
+        shard_count=shard_count,
+    )
+    controller.start()
+    conn.send("ready")
+    try:
+        conn.recv()
+    except (EOFError, OSError):
+        pass
+    controller.stop()
+    store.close()
+
+
 def run_control_plane(
     conn,
     policy: str = "packed-fit",
@@ -25,9 +60,17 @@ def run_control_plane(
     workers: int = 4,
     port: int = 0,
     persist_path: Optional[str] = None,
+    controller_shards: int = 1,
 ) -> None:
-    """multiprocessing target: start store+controller, report the port over
-    `conn`, run until the parent sends anything (or closes the pipe)."""
+    """multiprocessing target: start store+controller(s), report the port
+    over `conn`, run until the parent sends anything (or closes the pipe).
+
+    controller_shards == 1 (default): the controller shares this process
+    with the store (in-process MemStore — lowest single-stream latency).
+    controller_shards > 1: this process serves the store only; K controller
+    shard PROCESSES connect over TCP, each owning pods by crc32 hash. Python
+    reconciles are GIL-bound, so shards are the scale-out axis for
+    many-agent clusters (one shard saturates near ~400 pods/s)."""
     # short GIL switch interval: the reconcile path is wakeup-latency bound
     sys.setswitchinterval(0.001)
     from instaslice_amd.controller.reconciler import PodController
@@ -36,19 +79,48 @@ def run_control_plane(
 
     store = MemStore(persist_path=persist_path)
     server = StoreServer(store=store, port=port).start()
-    controller = PodController(
-        store,
-        policy=policy,
-        teardown_grace_s=teardown_grace_s,
-        requeue_no_fit_s=requeue_no_fit_s,
-        workers=workers,
-    )
-    controller.start()
+    controller = None
+    shard_procs = []
+    if controller_shards <= 1:
+        controller = PodController(
+            store,
+            policy=policy,
+            teardown_grace_s=teardown_grace_s,
+            requeue_no_fit_s=requeue_no_fit_s,
+            workers=workers,
+        )
+        controller.start()
+    else:
+        import multiprocessing as mp
+
+        ctx = mp.get_context("spawn")
+        for i in range(controller_shards):
+            parent_c, child_c = ctx.Pipe()
+            p = ctx.Process(
+                target=_run_controller_shard,
+                args=(server.port, policy, teardown_grace_s, requeue_no_fit_s,
+                      2, i, controller_shards, child_c),
+                daemon=True, name=f"controller-shard-{i}",
+            )
+            p.start()
+            shard_procs.append((p, parent_c))
+        for _, c in shard_procs:
+            c.recv()  # wait until every shard's watches are live
     conn.send(server.port)
     try:
         conn.recv()  # blocks until shutdown request or EOF
     except (EOFError, OSError):
         pass
-    controller.stop()
+    if controller is not None:
+        controller.stop()
+    for p, c in shard_procs:
+        try:
+            c.send("stop")
+        except (BrokenPipeError, OSError):
+            pass
+    for p, _ in shard_procs:
+        p.join(timeout=3.0)
+        if p.is_alive():
+            p.terminate()
     server.stop()
     store.close()

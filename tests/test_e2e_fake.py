@@ -6,6 +6,8 @@ the complete creating -> created -> ungated -> deleted machine runs against
 a fake 8x MI355X node with no hardware.
 """
 
+import time
+
 import pytest
 
 from instaslice_amd.api.types import AllocationStatus
@@ -331,3 +333,57 @@ def test_gang_spills_when_gpu_full():
         assert sizes == [1, 8]  # one full GPU + one spill
     finally:
         c.stop()
+
+
+def test_sharded_controllers_partition_pod_ownership():
+    """Two controller shards against one store: every pod is handled by
+    exactly one shard (crc32 ownership), all pods schedule, no double
+    allocation. This is the scale-out axis for many-agent clusters
+    (runtime/controlplane.py); the reference runs exactly one controller."""
+    from instaslice_amd.agent.daemonset import NodeAgent
+    from instaslice_amd.api.types import new_pod
+    from instaslice_amd.controller.reconciler import PodController
+    from instaslice_amd.store.memstore import MemStore
+
+    store = MemStore()
+    shards = [
+        PodController(store, teardown_grace_s=0.0, shard_index=i,
+                      shard_count=2, workers=1)
+        for i in range(2)
+    ]
+    agent = NodeAgent(store, FakeAmdSmi(num_gpus=2, node_name="node-0"),
+                      "node-0", heartbeat_every_s=0)
+    agent.start()
+    for c in shards:
+        c.start()
+    try:
+        # both shards must own at least one pod from this name set
+        owned = {i: [] for i in range(2)}
+        for i in range(12):
+            name = f"sp-{i}"
+            for c in shards:
+                if c._owns("default", name):
+                    owned[c.shard_index].append(name)
+            store.create(new_pod(name, profile="cpx-1x36"))
+        assert owned[0] and owned[1], "hash didn't split this name set"
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            pods = store.list("Pod")
+            if pods and all(not p["spec"].get("schedulingGates") for p in pods):
+                break
+            time.sleep(0.05)
+        pods = store.list("Pod")
+        assert all(not p["spec"].get("schedulingGates") for p in pods)
+        cr = store.get("Instaslice", "node-0", "instaslice-system")
+        allocs = cr["spec"]["allocations"]
+        assert len(allocs) == 12
+        slots = {(a["gpuUUID"], a["ordinal"]) for a in allocs.values()}
+        assert len(slots) == 12, "duplicate slot handed out across shards"
+        # shard latency stats only cover owned pods
+        n0 = len(shards[0].alloc_latency_s)
+        n1 = len(shards[1].alloc_latency_s)
+        assert n0 == len(owned[0]) and n1 == len(owned[1])
+    finally:
+        for c in shards:
+            c.stop()
+        agent.stop()

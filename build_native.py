@@ -83,6 +83,15 @@ def build(force: bool = False) -> None:
     if force or not newer(out, [src, hdr]):
         run([HIPCC, f"--offload-arch={ARCH}", *common, src, "-o", out])
 
+    # 5. instaslice-stored: native store daemon (plain host C++, g++ —
+    # no ROCm dependency; runs on any node incl. the CPU test tier)
+    store_dir = ROOT / "instaslice_amd" / "store"
+    src = store_dir / "csrc" / "stored_main.cpp"
+    hdr2 = store_dir / "csrc" / "msgpack_value.hpp"
+    out = bin_dir / "instaslice-stored"
+    if force or not newer(out, [src, hdr2]):
+        run(["g++", "-O2", "-std=c++17", "-pthread", src, "-o", out])
+
     print("native build complete")
 
 

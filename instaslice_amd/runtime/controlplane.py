@@ -61,29 +61,50 @@ def run_control_plane(
     port: int = 0,
     persist_path: Optional[str] = None,
     controller_shards: int = 1,
+    native_store: Optional[bool] = None,
 ) -> None:
     """multiprocessing target: start store+controller(s), report the port
     over `conn`, run until the parent sends anything (or closes the pipe).
 
     controller_shards == 1 (default): the controller shares this process
-    with the store (in-process MemStore — lowest single-stream latency).
-    controller_shards > 1: this process serves the store only; K controller
-    shard PROCESSES connect over TCP, each owning pods by crc32 hash. Python
-    reconciles are GIL-bound, so shards are the scale-out axis for
-    many-agent clusters (one shard saturates near ~400 pods/s)."""
+    with the Python store (in-process MemStore — lowest single-stream
+    latency). controller_shards > 1: the store serves alone and K controller
+    shard PROCESSES connect over TCP, each owning pods by crc32 hash —
+    Python reconciles are GIL-bound, so shards are the scale-out axis
+    (one shard saturates near ~400 pods/s).
+
+    native_store (None = auto): run the C++ store daemon
+    (instaslice-stored) instead of the Python server. Auto picks it for
+    sharded mode when the binary is built and no persistence is requested —
+    the Python server's GIL-bound wire handling is the next ceiling once
+    controllers shard."""
     # short GIL switch interval: the reconcile path is wakeup-latency bound
     sys.setswitchinterval(0.001)
     from instaslice_amd.controller.reconciler import PodController
     from instaslice_amd.store.memstore import MemStore
-    from instaslice_amd.store.netstore import StoreServer
+    from instaslice_amd.store.native import NativeStoreServer, stored_available
+    from instaslice_amd.store.netstore import NetStoreClient, StoreServer
 
-    store = MemStore(persist_path=persist_path)
-    server = StoreServer(store=store, port=port).start()
+    if native_store is None:
+        # measured: the daemon beats the Python server even for a single
+        # rank WITH the controller paying TCP round-trips (p50 2.9 vs 4.9 ms
+        # on the dev box) — the Python server's wire handling costs more
+        # than the extra hops. Python remains for persistence mode.
+        native_store = persist_path is None and stored_available()
+    store = None
+    if native_store:
+        server = NativeStoreServer(port=port).start()
+    else:
+        store = MemStore(persist_path=persist_path)
+        server = StoreServer(store=store, port=port).start()
     controller = None
     shard_procs = []
+    ctl_client = None
     if controller_shards <= 1:
+        if store is None:
+            ctl_client = NetStoreClient("127.0.0.1", server.port)
         controller = PodController(
-            store,
+            store if store is not None else ctl_client,
             policy=policy,
             teardown_grace_s=teardown_grace_s,
             requeue_no_fit_s=requeue_no_fit_s,
@@ -122,5 +143,8 @@ def run_control_plane(
         p.join(timeout=3.0)
         if p.is_alive():
             p.terminate()
+    if ctl_client is not None:
+        ctl_client.close()
     server.stop()
-    store.close()
+    if store is not None:
+        store.close()

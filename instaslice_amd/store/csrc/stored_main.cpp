@@ -1,0 +1,709 @@
+// instaslice-stored: native store daemon for the instaslice-amd control
+// plane.
+//
+// Speaks exactly the protocol of store/netstore.py (4-byte big-endian length
+// + msgpack map per message; verbs create/get/list/update/delete/patch/
+// batch/watch/ping with quiet + watch filters + replay) against the object
+// semantics of store/memstore.py (resourceVersion optimistic concurrency,
+// finalizer/deletionTimestamp two-phase delete, PATCH op grammar, filtered
+// watches). The Python MemStore remains the reference implementation — the
+// CPU test tier runs the same battery against both (tests/test_native_store
+// .py) — but under load the daemon is the control plane's scale-out floor:
+// the Python server serializes ~1 ms of GIL-bound wire handling per pod
+// lifecycle, which caps an 8-agent cluster near ~400 pods/s; this server
+// handles each connection on its own thread with a short critical section
+// around the object map, and encodes each watch event once.
+//
+// Reference analog: none — the reference (project-codeflare/instaslice)
+// outsources state to the Kubernetes API server + etcd. This daemon is the
+// API-server double for standalone/bench deployments.
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <cstdio>
+#include <cstring>
+#include <deque>
+#include <map>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "msgpack_value.hpp"
+
+namespace stored {
+
+// ---- store errors ----------------------------------------------------------
+
+struct StoreError {
+  const char* type;  // "Conflict" | "NotFound" | "AlreadyExists" | "Error"
+  std::string msg;
+};
+
+// ---- watch plumbing --------------------------------------------------------
+
+struct Conn;  // fwd
+
+struct WatchSub {
+  Conn* conn;
+  int64_t watch_id;
+  bool has_filters = false;
+  Value filters;           // array of {kind,name,namespace,labels} maps
+  std::string kind;        // kind-only subscription ("" = all)
+  std::atomic<bool> dead{false};
+};
+
+// per-connection outbox: responses and watch events share one ordered queue
+// flushed by a dedicated writer thread, so a slow consumer never blocks the
+// store mutex or another connection
+struct Conn {
+  int fd;
+  std::mutex out_mu;
+  std::condition_variable out_cv;
+  std::deque<std::string> outbox;
+  bool closing = false;
+  std::vector<std::shared_ptr<WatchSub>> subs;
+
+  void enqueue(std::string frame) {
+    {
+      std::lock_guard<std::mutex> g(out_mu);
+      if (closing) return;
+      outbox.push_back(std::move(frame));
+    }
+    out_cv.notify_one();
+  }
+};
+
+inline std::string frame(const std::string& payload) {
+  std::string out;
+  out.reserve(payload.size() + 4);
+  uint32_t n = static_cast<uint32_t>(payload.size());
+  out.push_back(static_cast<char>(n >> 24));
+  out.push_back(static_cast<char>(n >> 16));
+  out.push_back(static_cast<char>(n >> 8));
+  out.push_back(static_cast<char>(n));
+  out += payload;
+  return out;
+}
+
+// ---- object store ----------------------------------------------------------
+
+class Store {
+ public:
+  // key = kind \x00 ns \x00 name (lexicographic == Python tuple sort)
+  static std::string key(const std::string& kind, const std::string& ns,
+                         const std::string& name) {
+    std::string k;
+    k.reserve(kind.size() + ns.size() + name.size() + 2);
+    k += kind;
+    k.push_back('\0');
+    k += ns;
+    k.push_back('\0');
+    k += name;
+    return k;
+  }
+
+  static std::string obj_key(const Value& obj) {
+    const Value* md = obj.find("metadata");
+    static const std::string kEmpty;
+    const std::string& kind = obj.str_or("kind", kEmpty);
+    const std::string& ns = md ? md->str_or("namespace", kEmpty) : kEmpty;
+    const std::string& name = md ? md->str_or("name", kEmpty) : kEmpty;
+    return key(kind, ns, name);
+  }
+
+  Value create(const Value& obj_in) {
+    std::vector<std::string> frames;
+    Value out;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      std::string k = obj_key(obj_in);
+      if (objects_.count(k))
+        throw StoreError{"AlreadyExists", k + " already exists"};
+      Value obj = deep_copy(obj_in);
+      bump_rv(obj);
+      objects_[k] = obj;
+      notify_locked("ADDED", obj, frames);
+      out = obj;  // shared internals are never mutated in place (snapshot-on-write)
+    }
+    flush(frames);
+    return out;
+  }
+
+  Value get(const std::string& kind, const std::string& ns, const std::string& name) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = objects_.find(key(kind, ns, name));
+    if (it == objects_.end())
+      throw StoreError{"NotFound", "(" + kind + ", " + ns + ", " + name + ") not found"};
+    return it->second;
+  }
+
+  Value list(const std::string& kind, const Value* ns_filter) {
+    std::lock_guard<std::mutex> g(mu_);
+    Value out = Value::arr();
+    std::string prefix = kind;
+    prefix.push_back('\0');
+    for (auto it = objects_.lower_bound(prefix); it != objects_.end(); ++it) {
+      if (it->first.compare(0, prefix.size(), prefix) != 0) break;
+      if (ns_filter && ns_filter->t == Value::T::Str) {
+        const Value* md = it->second.find("metadata");
+        static const std::string kEmpty;
+        if (!md || md->str_or("namespace", kEmpty) != ns_filter->s) continue;
+      }
+      out.a->push_back(it->second);
+    }
+    return out;
+  }
+
+  Value update(const Value& obj_in) {
+    std::vector<std::string> frames;
+    Value out;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      std::string k = obj_key(obj_in);
+      auto it = objects_.find(k);
+      if (it == objects_.end()) throw StoreError{"NotFound", k + " not found"};
+      Value& cur = it->second;
+      const Value* cur_md = cur.find("metadata");
+      const Value* new_md = obj_in.find("metadata");
+      const Value* sent_rv = new_md ? new_md->find("resourceVersion") : nullptr;
+      if (sent_rv && !sent_rv->is_nil()) {
+        const Value* cur_rv = cur_md ? cur_md->find("resourceVersion") : nullptr;
+        if (!cur_rv || !deep_equal(*sent_rv, *cur_rv))
+          throw StoreError{"Conflict", k + ": stale resourceVersion"};
+      }
+      Value obj = deep_copy(obj_in);
+      bump_rv(obj);
+      // deletionTimestamp is sticky (k8s semantics)
+      const Value* cur_dt = cur_md ? cur_md->find("deletionTimestamp") : nullptr;
+      Value* md = obj.find("metadata");
+      if (cur_dt && cur_dt->truthy()) {
+        const Value* new_dt = md ? md->find("deletionTimestamp") : nullptr;
+        if (!new_dt || !new_dt->truthy()) md->setkey("deletionTimestamp", *cur_dt);
+      }
+      out = commit_locked(k, std::move(obj), frames);
+    }
+    flush(frames);
+    return out;
+  }
+
+  void del(const std::string& kind, const std::string& ns, const std::string& name) {
+    std::vector<std::string> frames;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      std::string k = key(kind, ns, name);
+      auto it = objects_.find(k);
+      if (it == objects_.end())
+        throw StoreError{"NotFound", "(" + kind + ", " + ns + ", " + name + ") not found"};
+      Value obj = it->second;
+      Value* md = obj.find("metadata");
+      const Value* fin = md ? md->find("finalizers") : nullptr;
+      if (fin && fin->truthy()) {
+        const Value* dt = md->find("deletionTimestamp");
+        if (!dt || !dt->truthy()) {
+          obj = deep_copy(obj);
+          md = obj.find("metadata");
+          double now = std::chrono::duration<double>(
+                           std::chrono::system_clock::now().time_since_epoch())
+                           .count();
+          md->setkey("deletionTimestamp", Value::real(now));
+          bump_rv(obj);
+          objects_[k] = obj;
+          notify_locked("MODIFIED", obj, frames);
+        }
+      } else {
+        objects_.erase(it);
+        notify_locked("DELETED", obj, frames);
+      }
+    }
+    flush(frames);
+  }
+
+  Value patch(const std::string& kind, const std::string& ns,
+              const std::string& name, const Value& ops) {
+    std::vector<std::string> frames;
+    Value out;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      std::string k = key(kind, ns, name);
+      auto it = objects_.find(k);
+      if (it == objects_.end())
+        throw StoreError{"NotFound", "(" + kind + ", " + ns + ", " + name + ") not found"};
+      Value obj = deep_copy(it->second);
+      if (ops.is_arr())
+        for (const auto& op : *ops.a) apply_op(obj, op);
+      bump_rv(obj);
+      out = commit_locked(k, std::move(obj), frames);
+    }
+    flush(frames);
+    return out;
+  }
+
+  // watch registration: replay + subscribe atomically under the store lock
+  void add_watch(std::shared_ptr<WatchSub> sub, bool replay) {
+    std::vector<std::string> frames;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      if (replay) {
+        for (const auto& kv : objects_) {
+          if (matches(*sub, kv.second)) {
+            frames.push_back(event_frame(sub->watch_id, "ADDED", kv.second));
+          }
+        }
+      }
+      watches_.push_back(sub);
+    }
+    for (auto& f : frames) sub->conn->enqueue(std::move(f));
+  }
+
+  void drop_conn_watches(Conn* c) {
+    std::lock_guard<std::mutex> g(mu_);
+    watches_.erase(
+        std::remove_if(watches_.begin(), watches_.end(),
+                       [c](const std::shared_ptr<WatchSub>& w) {
+                         return w->conn == c || w->dead.load();
+                       }),
+        watches_.end());
+  }
+
+ private:
+  void bump_rv(Value& obj) {
+    ++rv_;
+    Value* md = obj.find("metadata");
+    if (!md) md = &obj.setkey("metadata", Value::map());
+    md->setkey("resourceVersion", Value::str(std::to_string(rv_)));
+  }
+
+  // shared commit tail for update/patch: store, handle finalizer-free
+  // deletion, notify
+  Value commit_locked(const std::string& k, Value obj,
+                      std::vector<std::string>& frames) {
+    const Value* md = obj.find("metadata");
+    const Value* dt = md ? md->find("deletionTimestamp") : nullptr;
+    const Value* fin = md ? md->find("finalizers") : nullptr;
+    if (dt && dt->truthy() && (!fin || !fin->truthy())) {
+      objects_.erase(k);
+      notify_locked("DELETED", obj, frames);
+    } else {
+      objects_[k] = obj;
+      notify_locked("MODIFIED", obj, frames);
+    }
+    return obj;
+  }
+
+  // PATCH op grammar — mirror of memstore.apply_patch_ops
+  void apply_op(Value& obj, const Value& op) {
+    const Value* pathv = op.find("path");
+    const Value* opname = op.find("op");
+    if (!pathv || !pathv->is_arr() || !opname || !opname->is_str())
+      throw StoreError{"Error", "malformed patch op"};
+    const auto& path = *pathv->a;
+    const std::string& kind_op = opname->s;
+
+    if (kind_op == "test") {
+      const Value* node = &obj;
+      bool missing = false;
+      for (const auto& pseg : path) {
+        if (!pseg.is_str() || !node->is_map()) { missing = true; break; }
+        node = node->find(pseg.s);
+        if (!node) { missing = true; break; }
+      }
+      const Value* absent = op.find("absent");
+      if (absent && absent->truthy()) {
+        if (!missing) throw StoreError{"Conflict", "patch test: expected absent"};
+      } else {
+        const Value* want = op.find("value");
+        static const Value kNil;
+        if (missing || !deep_equal(*node, want ? *want : kNil))
+          throw StoreError{"Conflict", "patch test: value mismatch"};
+      }
+      return;
+    }
+
+    if (path.empty()) throw StoreError{"Error", "empty patch path"};
+    // navigate to parent, creating dicts along the way
+    Value* node = &obj;
+    for (size_t k = 0; k + 1 < path.size(); ++k) {
+      if (!path[k].is_str()) throw StoreError{"Error", "non-string path"};
+      Value* nxt = node->is_map() ? node->find(path[k].s) : nullptr;
+      if (!nxt || (!nxt->is_map() && !nxt->is_arr()))
+        nxt = &node->setkey(path[k].s, Value::map());
+      node = nxt;
+    }
+    if (!path.back().is_str()) throw StoreError{"Error", "non-string path leaf"};
+    const std::string& leaf = path.back().s;
+    const Value* val = op.find("value");
+    static const Value kNil;
+    const Value& v = val ? *val : kNil;
+
+    if (kind_op == "set") {
+      node->setkey(leaf, deep_copy(v));
+    } else if (kind_op == "merge") {
+      Value* tgt = node->find(leaf);
+      if (!tgt || !tgt->is_map()) tgt = &node->setkey(leaf, Value::map());
+      if (v.is_map())
+        for (const auto& kv : *v.m) tgt->setkey(kv.first, deep_copy(kv.second));
+    } else if (kind_op == "delete") {
+      node->erase(leaf);
+    } else if (kind_op == "add_to_set") {
+      Value* cur = node->find(leaf);
+      Value list = (cur && cur->is_arr()) ? deep_copy(*cur) : Value::arr();
+      bool present = false;
+      for (const auto& e : *list.a)
+        if (deep_equal(e, v)) { present = true; break; }
+      if (!present) {
+        list.a->push_back(deep_copy(v));
+        std::sort(list.a->begin(), list.a->end(), [](const Value& x, const Value& y) {
+          if (x.t == Value::T::Int && y.t == Value::T::Int) return x.i < y.i;
+          if (x.is_str() && y.is_str()) return x.s < y.s;
+          double xv = x.t == Value::T::Int ? static_cast<double>(x.i) : x.f;
+          double yv = y.t == Value::T::Int ? static_cast<double>(y.i) : y.f;
+          return xv < yv;
+        });
+      }
+      node->setkey(leaf, std::move(list));
+    } else if (kind_op == "remove_from_set") {
+      Value* cur = node->find(leaf);
+      if (cur && cur->is_arr()) {
+        Value list = Value::arr();
+        for (const auto& e : *cur->a)
+          if (!deep_equal(e, v)) list.a->push_back(deep_copy(e));
+        node->setkey(leaf, std::move(list));
+      }
+    } else {
+      throw StoreError{"Error", "unknown patch op " + kind_op};
+    }
+  }
+
+  static bool filter_matches(const Value& f, const Value& obj) {
+    static const std::string kEmpty;
+    const Value* want_kind = f.find("kind");
+    if (want_kind && want_kind->is_str() &&
+        want_kind->s != obj.str_or("kind", kEmpty))
+      return false;
+    const Value* md = obj.find("metadata");
+    const Value* want_name = f.find("name");
+    if (want_name && want_name->is_str()) {
+      if (!md || md->str_or("name", kEmpty) != want_name->s) return false;
+    }
+    const Value* want_ns = f.find("namespace");
+    if (want_ns && want_ns->is_str()) {
+      if (!md || md->str_or("namespace", kEmpty) != want_ns->s) return false;
+    }
+    const Value* want_labels = f.find("labels");
+    if (want_labels && want_labels->is_map() && !want_labels->m->empty()) {
+      const Value* have = md ? md->find("labels") : nullptr;
+      for (const auto& kv : *want_labels->m) {
+        const Value* hv = have ? have->find(kv.first) : nullptr;
+        if (!hv || !deep_equal(*hv, kv.second)) return false;
+      }
+    }
+    return true;
+  }
+
+  static bool matches(const WatchSub& w, const Value& obj) {
+    if (w.has_filters) {
+      for (const auto& f : *w.filters.a)
+        if (filter_matches(f, obj)) return true;
+      return false;
+    }
+    static const std::string kEmpty;
+    return w.kind.empty() || w.kind == obj.str_or("kind", kEmpty);
+  }
+
+  // encode the event payload ONCE; per-watch frames only prepend the header
+  static std::string event_frame(int64_t wid, const char* type, const Value& obj) {
+    Value ev = Value::map();
+    ev.setkey("watch_id", Value::integer(wid));
+    Value pair = Value::arr();
+    pair.a->push_back(Value::str(type));
+    pair.a->push_back(obj);  // shallow share is fine: pack() only reads
+    ev.setkey("event", std::move(pair));
+    std::string payload;
+    pack(ev, payload);
+    return frame(payload);
+  }
+
+  void notify_locked(const char* type, const Value& obj,
+                     std::vector<std::string>& frames_out) {
+    // pack [type, obj] once, splice per-watch headers around it
+    std::string ev_payload;
+    {
+      Value pair = Value::arr();
+      pair.a->push_back(Value::str(type));
+      pair.a->push_back(obj);
+      pack(pair, ev_payload);
+    }
+    bool any_dead = false;
+    for (const auto& w : watches_) {
+      if (w->dead.load()) { any_dead = true; continue; }
+      if (!matches(*w, obj)) continue;
+      Value hdr = Value::map();  // {"watch_id": N, "event": <spliced>}
+      std::string payload;
+      payload.push_back(static_cast<char>(0x82));
+      pack_str("watch_id", payload);
+      pack_int(w->watch_id, payload);
+      pack_str("event", payload);
+      payload += ev_payload;
+      w->conn->enqueue(frame(payload));
+      (void)frames_out;
+    }
+    if (any_dead) {
+      watches_.erase(std::remove_if(watches_.begin(), watches_.end(),
+                                    [](const std::shared_ptr<WatchSub>& w) {
+                                      return w->dead.load();
+                                    }),
+                     watches_.end());
+    }
+  }
+
+  void flush(std::vector<std::string>&) {}  // events already enqueued
+
+  std::mutex mu_;
+  std::map<std::string, Value> objects_;
+  uint64_t rv_ = 0;
+  std::vector<std::shared_ptr<WatchSub>> watches_;
+};
+
+// ---- per-connection handling ----------------------------------------------
+
+Value ok_response(const Value* rid, Value result) {
+  Value resp = Value::map();
+  if (rid) resp.setkey("id", *rid);
+  resp.setkey("ok", Value::boolean(true));
+  resp.setkey("result", std::move(result));
+  return resp;
+}
+
+Value err_response(const Value* rid, const StoreError& e) {
+  Value resp = Value::map();
+  if (rid) resp.setkey("id", *rid);
+  resp.setkey("ok", Value::boolean(false));
+  Value err = Value::map();
+  err.setkey("type", Value::str(e.type));
+  err.setkey("msg", Value::str(e.msg));
+  resp.setkey("error", std::move(err));
+  return resp;
+}
+
+Value execute(Store& store, const Value& req) {
+  static const std::string kEmpty;
+  const std::string& verb = req.str_or("verb", kEmpty);
+  const std::string& kind = req.str_or("kind", kEmpty);
+  const std::string& name = req.str_or("name", kEmpty);
+  const std::string& ns = req.str_or("namespace", kEmpty);
+  if (verb == "create") {
+    const Value* obj = req.find("obj");
+    if (!obj) throw StoreError{"Error", "create: missing obj"};
+    return store.create(*obj);
+  }
+  if (verb == "get") return store.get(kind, ns, name);
+  if (verb == "list") return store.list(kind, req.find("namespace"));
+  if (verb == "update") {
+    const Value* obj = req.find("obj");
+    if (!obj) throw StoreError{"Error", "update: missing obj"};
+    return store.update(*obj);
+  }
+  if (verb == "delete") {
+    store.del(kind, ns, name);
+    return Value::nil();
+  }
+  if (verb == "patch") {
+    const Value* ops = req.find("ops");
+    static const Value kNilOps;
+    return store.patch(kind, ns, name, ops ? *ops : kNilOps);
+  }
+  if (verb == "ping") return Value::str("pong");
+  throw StoreError{"Error", "unknown verb '" + verb + "'"};
+}
+
+void writer_loop(std::shared_ptr<Conn> conn) {
+  for (;;) {
+    std::string data;
+    {
+      std::unique_lock<std::mutex> g(conn->out_mu);
+      conn->out_cv.wait(g, [&] { return conn->closing || !conn->outbox.empty(); });
+      if (conn->closing && conn->outbox.empty()) return;
+      // coalesce queued frames into one send
+      size_t total = 0;
+      while (!conn->outbox.empty() && total < (1u << 20)) {
+        total += conn->outbox.front().size();
+        data += conn->outbox.front();
+        conn->outbox.pop_front();
+      }
+    }
+    const char* p = data.data();
+    size_t left = data.size();
+    while (left > 0) {
+      ssize_t n = ::send(conn->fd, p, left, MSG_NOSIGNAL);
+      if (n <= 0) {
+        std::lock_guard<std::mutex> g(conn->out_mu);
+        conn->closing = true;
+        conn->outbox.clear();
+        return;
+      }
+      p += n;
+      left -= static_cast<size_t>(n);
+    }
+  }
+}
+
+bool read_exact(int fd, char* buf, size_t n) {
+  size_t got = 0;
+  while (got < n) {
+    ssize_t r = ::recv(fd, buf + got, n - got, 0);
+    if (r <= 0) return false;
+    got += static_cast<size_t>(r);
+  }
+  return true;
+}
+
+void reader_loop(Store& store, std::shared_ptr<Conn> conn) {
+  std::thread writer(writer_loop, conn);
+  int64_t next_watch_id = 0;
+  for (;;) {
+    char hdr[4];
+    if (!read_exact(conn->fd, hdr, 4)) break;
+    uint32_t len = (static_cast<uint8_t>(hdr[0]) << 24) |
+                   (static_cast<uint8_t>(hdr[1]) << 16) |
+                   (static_cast<uint8_t>(hdr[2]) << 8) |
+                   static_cast<uint8_t>(hdr[3]);
+    if (len > (64u << 20)) break;  // oversized frame: drop connection
+    std::string buf(len, '\0');
+    if (!read_exact(conn->fd, buf.data(), len)) break;
+
+    Value req;
+    try {
+      req = unpack(buf);
+    } catch (const std::exception&) {
+      break;  // protocol desync: close
+    }
+    const Value* rid = req.find("id");
+    static const std::string kEmpty;
+    const std::string& verb = req.str_or("verb", kEmpty);
+    const Value* quietv = req.find("quiet");
+    bool quiet = quietv && quietv->truthy();
+
+    Value resp;
+    try {
+      if (verb == "watch") {
+        auto sub = std::make_shared<WatchSub>();
+        sub->conn = conn.get();
+        sub->watch_id = ++next_watch_id;
+        const Value* filters = req.find("filters");
+        if (filters && filters->is_arr()) {
+          sub->has_filters = true;
+          sub->filters = *filters;
+        } else {
+          const Value* kindv = req.find("kind");
+          sub->kind = (kindv && kindv->is_str()) ? kindv->s : "";
+        }
+        const Value* replayv = req.find("replay");
+        bool replay = !replayv || replayv->truthy();
+        // respond BEFORE replay events hit the outbox? Python server sends
+        // the watch_id response first; preserve that order.
+        Value result = Value::map();
+        result.setkey("watch_id", Value::integer(sub->watch_id));
+        std::string payload;
+        pack(ok_response(rid, std::move(result)), payload);
+        conn->enqueue(frame(payload));
+        conn->subs.push_back(sub);
+        store.add_watch(sub, replay);
+        continue;
+      }
+      if (verb == "batch") {
+        const Value* reqs = req.find("requests");
+        Value results = Value::arr();
+        if (reqs && reqs->is_arr()) {
+          for (const auto& sub_req : *reqs->a) {
+            Value entry = Value::map();
+            try {
+              Value r = execute(store, sub_req);
+              entry.setkey("ok", Value::boolean(true));
+              entry.setkey("result", quiet ? Value::nil() : std::move(r));
+            } catch (const StoreError& e) {
+              entry.setkey("ok", Value::boolean(false));
+              entry.setkey("result", Value::nil());
+              Value err = Value::map();
+              err.setkey("type", Value::str(e.type));
+              err.setkey("msg", Value::str(e.msg));
+              entry.setkey("error", std::move(err));
+            }
+            results.a->push_back(std::move(entry));
+          }
+        }
+        resp = ok_response(rid, std::move(results));
+      } else {
+        Value r = execute(store, req);
+        resp = ok_response(rid, quiet ? Value::nil() : std::move(r));
+      }
+    } catch (const StoreError& e) {
+      resp = err_response(rid, e);
+    } catch (const std::exception& e) {
+      resp = err_response(rid, StoreError{"Error", e.what()});
+    }
+    std::string payload;
+    pack(resp, payload);
+    conn->enqueue(frame(payload));
+  }
+
+  // teardown
+  for (auto& s : conn->subs) s->dead.store(true);
+  store.drop_conn_watches(conn.get());
+  {
+    std::lock_guard<std::mutex> g(conn->out_mu);
+    conn->closing = true;
+  }
+  conn->out_cv.notify_one();
+  writer.join();
+  ::close(conn->fd);
+}
+
+}  // namespace stored
+
+int main(int argc, char** argv) {
+  using namespace stored;
+  int port = 0;
+  if (argc > 1) port = std::atoi(argv[1]);
+
+  int srv = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (srv < 0) { perror("socket"); return 1; }
+  int one = 1;
+  ::setsockopt(srv, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  sockaddr_in addr{};
+  addr.sin_family = AF_INET;
+  addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+  addr.sin_port = htons(static_cast<uint16_t>(port));
+  if (::bind(srv, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)) != 0) {
+    perror("bind");
+    return 1;
+  }
+  socklen_t alen = sizeof(addr);
+  ::getsockname(srv, reinterpret_cast<sockaddr*>(&addr), &alen);
+  if (::listen(srv, 64) != 0) { perror("listen"); return 1; }
+  // parent (store/native.py) parses this line for the chosen port
+  std::printf("LISTENING %d\n", ntohs(addr.sin_port));
+  std::fflush(stdout);
+
+  Store store;
+  for (;;) {
+    int fd = ::accept(srv, nullptr, nullptr);
+    if (fd < 0) {
+      if (errno == EINTR) continue;
+      break;
+    }
+    ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    auto conn = std::make_shared<Conn>();
+    conn->fd = fd;
+    std::thread(reader_loop, std::ref(store), conn).detach();
+  }
+  return 0;
+}

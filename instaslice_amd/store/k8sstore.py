@@ -318,9 +318,9 @@ class K8sStore:
 
     def update_with_retry(
         self, kind: str, name: str, namespace: str,
-        mutate: Callable[[dict], Optional[dict]], attempts: int = 10,
+        mutate: Callable[[dict], Optional[dict]], attempts: int = 25,
     ):
-        for _ in range(attempts):
+        for attempt in range(attempts):
             try:
                 obj = self.get(kind, name, namespace)
             except NotFound:
@@ -331,5 +331,10 @@ class K8sStore:
             try:
                 return self.update(new)
             except Conflict:
+                if attempt >= 2:  # hot object: jittered backoff breaks livelock
+                    import random
+                    import time as _time
+
+                    _time.sleep(random.random() * 0.002 * attempt)
                 continue
         raise Conflict(f"update_with_retry exhausted for {kind}/{name}")

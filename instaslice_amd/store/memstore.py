@@ -450,7 +450,7 @@ class MemStore:
         """Get-mutate-update loop absorbing Conflicts. `mutate` returns the
         modified object or None to abort. This is the pattern the reference
         implements as requeue-on-conflict (instaslice_controller.go:93)."""
-        for _ in range(attempts):
+        for attempt in range(attempts):
             try:
                 obj = self.get(kind, name, namespace)
             except NotFound:
@@ -461,5 +461,10 @@ class MemStore:
             try:
                 return self.update(new)
             except Conflict:
+                if attempt >= 2:  # hot object: jittered backoff breaks livelock
+                    import random
+                    import time as _time
+
+                    _time.sleep(random.random() * 0.002 * attempt)
                 continue
         raise Conflict(f"update_with_retry: {attempts} attempts exhausted for {kind}/{name}")

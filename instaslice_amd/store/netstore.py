@@ -329,9 +329,9 @@ class NetStoreClient:
 
     def update_with_retry(
         self, kind: str, name: str, namespace: str,
-        mutate: Callable[[dict], Optional[dict]], attempts: int = 10,
+        mutate: Callable[[dict], Optional[dict]], attempts: int = 25,
     ) -> Optional[dict]:
-        for _ in range(attempts):
+        for attempt in range(attempts):
             try:
                 obj = self.get(kind, name, namespace)
             except NotFound:
@@ -342,6 +342,11 @@ class NetStoreClient:
             try:
                 return self.update(new)
             except Conflict:
+                if attempt >= 2:  # hot object: jittered backoff breaks livelock
+                    import random
+                    import time as _time
+
+                    _time.sleep(random.random() * 0.002 * attempt)
                 continue
         raise Conflict(f"update_with_retry: {attempts} attempts exhausted for {kind}/{name}")
 

@@ -1,0 +1,286 @@
+"""Protocol-parity battery: the C++ store daemon (instaslice-stored) must be
+indistinguishable from MemStore behind NetStoreClient. Every scenario runs
+against BOTH backends via the parametrized fixture; the native daemon is the
+control plane's scale-out path (store/csrc/stored_main.cpp), MemStore the
+reference semantics (store/memstore.py)."""
+
+import threading
+import time
+
+import pytest
+
+from instaslice_amd.api.types import new_pod
+from instaslice_amd.store import AlreadyExists, Conflict, MemStore, NotFound
+from instaslice_amd.store.native import NativeStoreServer, stored_available
+from instaslice_amd.store.netstore import NetStoreClient, StoreServer
+
+needs_stored = pytest.mark.skipif(
+    not stored_available(),
+    reason="instaslice-stored not built (python build_native.py)",
+)
+
+
+@pytest.fixture(params=["python", "native"])
+def client(request):
+    if request.param == "native":
+        if not stored_available():
+            pytest.skip("instaslice-stored not built")
+        server = NativeStoreServer().start()
+    else:
+        server = StoreServer().start()
+    c = NetStoreClient("127.0.0.1", server.port)
+    yield c
+    c.close()
+    server.stop()
+
+
+def _obj(name, kind="Thing", ns="", **extra):
+    return {"apiVersion": "v1", "kind": kind,
+            "metadata": {"name": name, "namespace": ns}, **extra}
+
+
+def test_crud_roundtrip(client):
+    client.create(_obj("a", x=1))
+    got = client.get("Thing", "a")
+    assert got["x"] == 1 and got["metadata"]["resourceVersion"] == "1"
+    with pytest.raises(AlreadyExists):
+        client.create(_obj("a"))
+    got["x"] = 2
+    updated = client.update(got)
+    assert updated["x"] == 2
+    assert updated["metadata"]["resourceVersion"] != "1"
+    with pytest.raises(Conflict):
+        client.update(got)  # stale rv
+    assert [o["metadata"]["name"] for o in client.list("Thing")] == ["a"]
+    client.delete("Thing", "a")
+    with pytest.raises(NotFound):
+        client.get("Thing", "a")
+    with pytest.raises(NotFound):
+        client.delete("Thing", "a")
+
+
+def test_list_sorted_and_namespaced(client):
+    client.create(_obj("b", ns="n2"))
+    client.create(_obj("a", ns="n1"))
+    client.create(_obj("c", kind="Other"))
+    names = [o["metadata"]["name"] for o in client.list("Thing")]
+    assert names == ["a", "b"]
+    assert [o["metadata"]["name"] for o in client.list("Thing", "n2")] == ["b"]
+
+
+def test_finalizer_two_phase_delete(client):
+    obj = _obj("f")
+    obj["metadata"]["finalizers"] = ["org.instaslice/accelarator"]
+    client.create(obj)
+    client.delete("Thing", "f")
+    got = client.get("Thing", "f")  # still there, deletionTimestamp set
+    assert got["metadata"]["deletionTimestamp"]
+    got["metadata"]["finalizers"] = []
+    client.update(got)  # finalizer removed -> object actually goes away
+    with pytest.raises(NotFound):
+        client.get("Thing", "f")
+
+
+def test_patch_ops(client):
+    client.create(_obj("p", spec={"allocations": {}, "n": 1}))
+    res = client.patch("Thing", "p", "", [
+        {"op": "set", "path": ["spec", "allocations", "u1", "allocationStatus"],
+         "value": "creating"},
+        {"op": "merge", "path": ["spec", "prepared"], "value": {"x": {"k": 1}}},
+        {"op": "add_to_set", "path": ["spec", "used"], "value": 3},
+        {"op": "add_to_set", "path": ["spec", "used"], "value": 1},
+        {"op": "add_to_set", "path": ["spec", "used"], "value": 3},
+    ])
+    assert res["spec"]["allocations"]["u1"]["allocationStatus"] == "creating"
+    assert res["spec"]["prepared"] == {"x": {"k": 1}}
+    assert res["spec"]["used"] == [1, 3]
+    res = client.patch("Thing", "p", "", [
+        {"op": "test", "path": ["spec", "allocations", "u1", "allocationStatus"],
+         "value": "creating"},
+        {"op": "set", "path": ["spec", "allocations", "u1", "allocationStatus"],
+         "value": "created"},
+        {"op": "remove_from_set", "path": ["spec", "used"], "value": 3},
+        {"op": "delete", "path": ["spec", "n"]},
+    ])
+    assert res["spec"]["allocations"]["u1"]["allocationStatus"] == "created"
+    assert res["spec"]["used"] == [1]
+    assert "n" not in res["spec"]
+    # failed test leaves the object untouched
+    before = client.get("Thing", "p")
+    with pytest.raises(Conflict):
+        client.patch("Thing", "p", "", [
+            {"op": "test", "path": ["spec", "allocations", "u1", "allocationStatus"],
+             "value": "creating"},
+            {"op": "set", "path": ["spec", "boom"], "value": True},
+        ])
+    after = client.get("Thing", "p")
+    assert after == before
+    with pytest.raises(Conflict):
+        client.patch("Thing", "p", "", [
+            {"op": "test", "path": ["spec", "allocations", "u1"], "absent": True},
+        ])
+    with pytest.raises(NotFound):
+        client.patch("Thing", "nope", "", [{"op": "set", "path": ["x"], "value": 1}])
+
+
+def test_batch_and_quiet(client):
+    res = client.batch([
+        {"verb": "create", "obj": _obj("b1")},
+        {"verb": "create", "obj": _obj("b1")},  # dup -> AlreadyExists
+        {"verb": "patch", "kind": "Thing", "name": "b1", "namespace": "",
+         "ops": [{"op": "set", "path": ["spec", "v"], "value": 7}]},
+        {"verb": "delete", "kind": "Thing", "name": "missing", "namespace": ""},
+    ])
+    assert [r["ok"] for r in res] == [True, False, True, False]
+    assert res[1]["error"]["type"] == "AlreadyExists"
+    assert res[2]["result"]["spec"]["v"] == 7
+    assert res[3]["error"]["type"] == "NotFound"
+    res = client.batch([
+        {"verb": "patch", "kind": "Thing", "name": "b1", "namespace": "",
+         "ops": [{"op": "set", "path": ["spec", "v"], "value": 8}]},
+    ], quiet=True)
+    assert res[0]["ok"] and res[0]["result"] is None
+    assert client.get("Thing", "b1")["spec"]["v"] == 8
+    assert client.patch("Thing", "b1", "",
+                        [{"op": "set", "path": ["spec", "v"], "value": 9}],
+                        quiet=True) is None
+
+
+def test_watch_kinds_replay_and_filters(client):
+    client.create(new_pod("w1", labels={"grp": "a"}))
+    client.create(new_pod("w2", labels={"grp": "b"}))
+    w = client.watch("Pod")  # replay
+    seen = {w.next(timeout=2)[1]["metadata"]["name"] for _ in range(2)}
+    assert seen == {"w1", "w2"}
+    wf = client.watch(replay=True, filters=[
+        {"kind": "Pod", "labels": {"grp": "a"}},
+        {"kind": "Thing", "name": "t9"},
+    ])
+    ev = wf.next(timeout=2)
+    assert ev[1]["metadata"]["name"] == "w1"
+    client.create(_obj("t8"))   # not matched
+    client.create(_obj("t9"))   # matched by name filter
+    ev = wf.next(timeout=2)
+    assert ev[0] == "ADDED" and ev[1]["metadata"]["name"] == "t9"
+    client.create(new_pod("w3", labels={"grp": "a"}))
+    ev = wf.next(timeout=2)
+    assert ev[1]["metadata"]["name"] == "w3"
+    w.stop()
+    wf.stop()
+
+
+def test_watch_modify_delete_events(client):
+    w = client.watch("Thing", replay=False)
+    client.create(_obj("m"))
+    obj = client.get("Thing", "m")
+    obj["x"] = 1
+    client.update(obj)
+    client.delete("Thing", "m")
+    events = [w.next(timeout=2) for _ in range(3)]
+    assert [e[0] for e in events] == ["ADDED", "MODIFIED", "DELETED"]
+    w.stop()
+
+
+def test_update_with_retry_conflict_absorption(client):
+    client.create(_obj("r", n=0))
+
+    def bump():
+        for _ in range(20):
+            client.update_with_retry(
+                "Thing", "r", "",
+                lambda o: {**o, "n": o["n"] + 1},
+            )
+
+    threads = [threading.Thread(target=bump) for _ in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert client.get("Thing", "r")["n"] == 80
+
+
+@needs_stored
+def test_native_concurrent_clients_and_patch_atomicity():
+    """Hammer one object with concurrent add_to_set patches from several
+    clients: every element must land exactly once (store-side atomicity)."""
+    server = NativeStoreServer().start()
+    clients = [NetStoreClient("127.0.0.1", server.port) for _ in range(4)]
+    try:
+        clients[0].create(_obj("hot", spec={"s": []}))
+
+        def worker(ci, base):
+            for k in range(25):
+                clients[ci].patch("Thing", "hot", "", [
+                    {"op": "add_to_set", "path": ["spec", "s"],
+                     "value": base * 100 + k},
+                ], quiet=True)
+
+        threads = [threading.Thread(target=worker, args=(i, i)) for i in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        got = clients[0].get("Thing", "hot")["spec"]["s"]
+        assert len(got) == 100 and got == sorted(got)
+    finally:
+        for c in clients:
+            c.close()
+        server.stop()
+
+
+@needs_stored
+def test_native_full_pod_lifecycle_stack():
+    """controller + agent + fake SMI against the NATIVE store daemon: the
+    complete creating -> created -> ungated -> deleted machine (the bench
+    topology with the C++ data plane)."""
+    from instaslice_amd.agent.daemonset import NodeAgent
+    from instaslice_amd.controller.reconciler import PodController
+    from instaslice_amd.smi.fake import FakeAmdSmi
+
+    server = NativeStoreServer().start()
+    cstore = NetStoreClient("127.0.0.1", server.port)
+    astore = NetStoreClient("127.0.0.1", server.port)
+    controller = PodController(cstore, teardown_grace_s=0.0, workers=2)
+    controller.requeue_no_fit_s = 0.05
+    agent = NodeAgent(astore, FakeAmdSmi(num_gpus=2, node_name="node-0"),
+                      "node-0")
+    agent.start()
+    controller.start()
+    bench = NetStoreClient("127.0.0.1", server.port)
+    try:
+        for i in range(10):
+            bench.create(new_pod(f"n{i}", profile="cpx-1x36"))
+        deadline = time.monotonic() + 20
+        while time.monotonic() < deadline:
+            pods = bench.list("Pod")
+            if len(pods) == 10 and all(
+                not p["spec"].get("schedulingGates") for p in pods
+            ):
+                break
+            time.sleep(0.05)
+        pods = bench.list("Pod")
+        assert all(not p["spec"].get("schedulingGates") for p in pods)
+        cr = bench.get("Instaslice", "node-0", "instaslice-system")
+        assert len(cr["spec"]["allocations"]) == 10
+        assert len(cr["spec"]["prepared"]) == 10
+        assert bench.get("ConfigMap", "n0", "default")["data"][
+            "ROCR_VISIBLE_DEVICES"]
+        # drain one pod completely
+        bench.delete("Pod", "n0", "default")
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            cr = bench.get("Instaslice", "node-0", "instaslice-system")
+            if not any(a["podName"] == "n0"
+                       for a in cr["spec"]["allocations"].values()):
+                break
+            time.sleep(0.05)
+        assert not any(a["podName"] == "n0"
+                       for a in cr["spec"]["allocations"].values())
+        with pytest.raises(NotFound):
+            bench.get("ConfigMap", "n0", "default")
+    finally:
+        controller.stop()
+        agent.stop()
+        for c in (cstore, astore, bench):
+            c.close()
+        server.stop()

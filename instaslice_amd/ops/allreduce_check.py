@@ -28,6 +28,10 @@ def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--numel", type=int, default=1 << 22)
     ap.add_argument("--iters", type=int, default=20)
+    ap.add_argument("--backend", choices=["auto", "gloo", "nccl"],
+                    default="auto",
+                    help="force a backend (the CPU test tier runs gloo "
+                         "even on a machine that has a GPU)")
     args = ap.parse_args()
 
     import torch
@@ -35,8 +39,12 @@ def main() -> int:
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
-    use_gpu = torch.cuda.is_available()
-    backend = "nccl" if use_gpu else "gloo"
+    if args.backend == "auto":
+        use_gpu = torch.cuda.is_available()
+        backend = "nccl" if use_gpu else "gloo"
+    else:
+        backend = args.backend
+        use_gpu = backend == "nccl"
 
     if world > 1:
         dist.init_process_group(backend=backend)

@@ -17,7 +17,7 @@ def _launch(nproc, env_extra=None):
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
          "--master-port", "29591", "-m", "instaslice_amd.ops.allreduce_check",
-         "--numel", "65536", "--iters", "5"],
+         "--numel", "65536", "--iters", "5", "--backend", "gloo"],
         capture_output=True, text=True, timeout=300, env=env, cwd=ROOT)
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
     line = [l for l in out.stdout.splitlines() if l.startswith("{")][0]

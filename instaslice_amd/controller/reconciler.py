@@ -27,7 +27,6 @@ from instaslice_amd.api.types import (
     AllocationStatus,
     pod_is_gated,
     pod_limits,
-    remove_finalizer,
     ungate_pod,
 )
 from instaslice_amd.controller.policy import AllocationPolicy, build_gpu_views, get_policy
@@ -206,29 +205,7 @@ class PodController:
                 group=group,
             )
 
-            def add_alloc(cr_obj: dict, a=alloc) -> Optional[dict]:
-                spec = cr_obj.setdefault("spec", {})
-                allocs = spec.setdefault("allocations", {})
-                if a.pod_uuid in allocs:
-                    return None  # raced with ourselves; done
-                # re-validate the placement against the fresh CR (another
-                # pod may have taken the ordinal between list() and now)
-                fresh_views = build_gpu_views(cr_obj["metadata"]["name"], spec)
-                for v in fresh_views:
-                    if v.uuid == a.gpu_uuid:
-                        if a.ordinal in v.occupied:
-                            return None  # lost the race; requeue will re-place
-                        if (
-                            v.compute_mode.value != a.compute_mode
-                            and v.occupied
-                        ):
-                            return None
-                allocs[a.pod_uuid] = a.to_dict()
-                return cr_obj
-
-            updated = self.store.update_with_retry(
-                "Instaslice", node_name, INSTASLICE_NS, add_alloc
-            )
+            updated = self._write_allocation(cr, node_name, alloc)
             if updated and md["uid"] in updated["spec"]["allocations"]:
                 self.log.debug(
                     "placed pod %s profile %s on %s/%s ordinal %d",
@@ -268,6 +245,55 @@ class PodController:
 
         self.store.update_with_retry(
             "Pod", md["name"], md.get("namespace", "default"), mut
+        )
+
+    def _write_allocation(self, cr: dict, node_name: str, alloc) -> Optional[dict]:
+        """Persist a placement decision. Fast path: one CAS patch — the
+        decision was taken on `cr`, so guard on its resourceVersion; any
+        concurrent CR write (another shard, an agent commit) fails the guard
+        and we fall back to the classic get-revalidate-update cycle, which
+        re-checks the ordinal/mode against the fresh object (reference's
+        placement race handling done atomically rather than by requeue,
+        instaslice_controller.go:93)."""
+        from instaslice_amd.store.memstore import Conflict
+
+        a = alloc
+        try:
+            return self.store.patch("Instaslice", node_name, INSTASLICE_NS, [
+                {"op": "test", "path": ["metadata", "resourceVersion"],
+                 "value": cr["metadata"]["resourceVersion"]},
+                {"op": "test", "path": ["spec", "allocations", a.pod_uuid],
+                 "absent": True},
+                {"op": "set", "path": ["spec", "allocations", a.pod_uuid],
+                 "value": a.to_dict()},
+            ])
+        except NotFound:
+            return None
+        except Conflict:
+            pass
+
+        def add_alloc(cr_obj: dict) -> Optional[dict]:
+            spec = cr_obj.setdefault("spec", {})
+            allocs = spec.setdefault("allocations", {})
+            if a.pod_uuid in allocs:
+                return None  # raced with ourselves; done
+            # re-validate the placement against the fresh CR (another
+            # pod may have taken the ordinal between list() and now)
+            fresh_views = build_gpu_views(cr_obj["metadata"]["name"], spec)
+            for v in fresh_views:
+                if v.uuid == a.gpu_uuid:
+                    if a.ordinal in v.occupied:
+                        return None  # lost the race; requeue will re-place
+                    if (
+                        v.compute_mode.value != a.compute_mode
+                        and v.occupied
+                    ):
+                        return None
+            allocs[a.pod_uuid] = a.to_dict()
+            return cr_obj
+
+        return self.store.update_with_retry(
+            "Instaslice", node_name, INSTASLICE_NS, add_alloc
         )
 
     def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str,
@@ -325,22 +351,41 @@ class PodController:
                 elapsed = time.time() - float(md["deletionTimestamp"])
                 if elapsed < self.teardown_grace_s:
                     return Result(requeue_after=self.teardown_grace_s - elapsed)
+            # ONE batched round-trip: mark the allocation deleted (agent
+            # tears it down on that event) AND strip our finalizer — the
+            # store drops the pod once no finalizers remain
+            from instaslice_amd import FINALIZER_NAME
+
+            reqs = []
             found = self._find_allocation(uid, self._crs())
             if found:
                 cr, alloc = found
                 if alloc["allocationStatus"] != AllocationStatus.DELETED:
-                    self._set_allocation_status(
-                        cr["metadata"]["name"], uid, AllocationStatus.DELETED
-                    )
-            # remove our finalizer; the store drops the pod when none remain
-            def strip(p: dict) -> Optional[dict]:
-                from instaslice_amd import FINALIZER_NAME
-
-                if FINALIZER_NAME not in (p["metadata"].get("finalizers") or []):
-                    return None
-                return remove_finalizer(p)
-
-            self.store.update_with_retry("Pod", name, namespace, strip)
+                    reqs.append(
+                        {"verb": "patch", "kind": "Instaslice",
+                         "name": cr["metadata"]["name"],
+                         "namespace": INSTASLICE_NS, "ops": [
+                             {"op": "test",
+                              "path": ["spec", "allocations", uid, "podUUID"],
+                              "value": uid},
+                             {"op": "set",
+                              "path": ["spec", "allocations", uid,
+                                       "allocationStatus"],
+                              "value": AllocationStatus.DELETED},
+                         ]})
+            if FINALIZER_NAME in (md.get("finalizers") or []):
+                # guard on our view: an unconditional strip would bump the
+                # rv (and emit MODIFIED) even when already stripped —
+                # an event loop for pods carrying foreign finalizers
+                reqs.append(
+                    {"verb": "patch", "kind": "Pod", "name": name,
+                     "namespace": namespace, "ops": [
+                         {"op": "remove_from_set",
+                          "path": ["metadata", "finalizers"],
+                          "value": FINALIZER_NAME},
+                     ]})
+            if reqs:
+                self.store.batch(reqs, quiet=True)
             return Result()
 
         if not pod_is_gated(pod):

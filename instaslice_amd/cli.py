@@ -50,12 +50,23 @@ def _wait_forever():
 
 def cmd_store(args) -> int:
     from instaslice_amd.store.memstore import MemStore
+    from instaslice_amd.store.native import NativeStoreServer, stored_available
     from instaslice_amd.store.netstore import StoreServer
 
-    backing = MemStore(persist_path=args.data) if args.data else None
-    server = StoreServer(store=backing, port=args.port).start()
-    log.info("store serving on %s:%d (data=%s)", server.host, server.port,
-             args.data or "in-memory")
+    backing = None
+    use_native = (args.native or (args.native is None and not args.data
+                                  and stored_available()))
+    if use_native:
+        if args.data:
+            raise SystemExit("--native store has no persistence; drop --data")
+        server = NativeStoreServer(port=args.port).start()
+        log.info("native store (instaslice-stored) serving on %s:%d",
+                 server.host, server.port)
+    else:
+        backing = MemStore(persist_path=args.data) if args.data else None
+        server = StoreServer(store=backing, port=args.port).start()
+        log.info("store serving on %s:%d (data=%s)", server.host, server.port,
+                 args.data or "in-memory")
     _wait_forever()
     server.stop()
     if backing:
@@ -76,7 +87,9 @@ def cmd_controller(args) -> int:
         log.info("waiting for leadership...")
         lease.wait_leader()
     controller = PodController(store, policy=args.policy,
-                               teardown_grace_s=args.grace)
+                               teardown_grace_s=args.grace,
+                               shard_index=args.shard_index,
+                               shard_count=args.shard_count)
     controller.start()
     srv = None
     if args.metrics_port:
@@ -198,7 +211,13 @@ def main(argv=None) -> int:
     p = sub.add_parser("store", help="run the state store server")
     p.add_argument("--port", type=int, default=7080)
     p.add_argument("--data", default=None,
-                   help="persist state to this JSON file (checkpoint/resume)")
+                   help="persist state to this JSON file (checkpoint/resume; "
+                        "forces the Python server)")
+    p.add_argument("--native", action="store_true", default=None,
+                   help="serve with the C++ instaslice-stored daemon "
+                        "(default: auto when built and --data is unset)")
+    p.add_argument("--no-native", dest="native", action="store_false",
+                   help="force the Python store server")
     p.set_defaults(fn=cmd_store)
 
     p = sub.add_parser("controller", help="run the cluster controller")
@@ -208,6 +227,10 @@ def main(argv=None) -> int:
                    help="teardown grace seconds (reference: 30)")
     p.add_argument("--metrics-port", type=int, default=8080)
     p.add_argument("--leader-elect", action="store_true")
+    p.add_argument("--shard-index", type=int, default=0,
+                   help="this controller's shard (crc32 pod ownership)")
+    p.add_argument("--shard-count", type=int, default=1,
+                   help="total controller shards (run one process per shard)")
     p.set_defaults(fn=cmd_controller)
 
     p = sub.add_parser("daemonset", help="run the per-node agent")

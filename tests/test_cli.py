@@ -96,3 +96,51 @@ def test_cli_end_to_end_processes():
             except subprocess.TimeoutExpired:
                 p.kill()
         server.stop()
+
+
+def test_cli_native_store_sharded_controllers():
+    """Native store daemon + 2 sharded controller processes + fake daemonset,
+    all via the CLI — the scale-out deployment shape."""
+    from instaslice_amd.store.native import NativeStoreServer, stored_available
+
+    if not stored_available():
+        pytest.skip("instaslice-stored not built")
+    server = NativeStoreServer().start()
+    addr = f"127.0.0.1:{server.port}"
+    procs = []
+
+    def spawn(*cmd):
+        p = subprocess.Popen([sys.executable, "-m", "instaslice_amd", *cmd],
+                             stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                             text=True)
+        procs.append(p)
+        return p
+
+    try:
+        for i in range(2):
+            spawn("controller", "--store", addr, "--grace", "0",
+                  "--metrics-port", "0", "--shard-index", str(i),
+                  "--shard-count", "2")
+        spawn("daemonset", "--store", addr, "--node-name", "cli-node",
+              "--fake", "2", "--metrics-port", "0")
+        # several pods so both shards own at least one with high likelihood
+        for k in range(6):
+            out = subprocess.run(
+                [sys.executable, "-m", "instaslice_amd", "submit", "--store",
+                 addr, "--name", f"shpod{k}", "--profile", "cpx-1x36",
+                 "--wait", "--timeout", "30"],
+                capture_output=True, text=True, timeout=60)
+            assert out.returncode == 0, out.stdout + out.stderr
+        client = NetStoreClient("127.0.0.1", server.port)
+        cr = client.get("Instaslice", "cli-node", "instaslice-system")
+        assert len(cr["spec"]["allocations"]) == 6
+        client.close()
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
+        server.stop()

@@ -85,6 +85,8 @@ class PodController:
         # pods currently marked unschedulable: re-reconciled on ANY node-state
         # change so freed capacity is picked up event-driven, not by polling
         self._unschedulable_keys: set = set()
+        # node -> (placements_dict, ProfileCatalog): see _profile_for
+        self._catalog_cache: dict = {}
         # workers > 1 is safe: the engine guarantees a key is never
         # reconciled concurrently (runtime/engine.py in-flight guard), and
         # placement races between different pods are caught by the
@@ -150,7 +152,15 @@ class PodController:
     def _profile_for(self, cr: dict, name: str) -> Optional[PartitionProfile]:
         placements = cr.get("spec", {}).get("placements") or {}
         if placements:
-            cat = ProfileCatalog.from_dict(placements)
+            # catalog parse is ~100 us and placements are static per node:
+            # cache per node, validated by dict equality (cheap vs re-parse)
+            node = cr["metadata"]["name"]
+            cached = self._catalog_cache.get(node)
+            if cached is not None and cached[0] == placements:
+                cat = cached[1]
+            else:
+                cat = ProfileCatalog.from_dict(placements)
+                self._catalog_cache[node] = (placements, cat)
             p = cat.by_name(name)
             if p:
                 return p

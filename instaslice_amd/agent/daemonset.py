@@ -316,6 +316,8 @@ class NodeAgent:
         ], quiet=True)
         if not res[0]["ok"] and res[0]["error"]["type"] != "AlreadyExists":
             self.log.warning("configmap create failed: %s", res[0]["error"])
+        if not res[1]["ok"]:
+            self.log.warning("node capacity pin failed: %s", res[1]["error"])
         commit = res[2]
         if not commit["ok"]:
             # Conflict: status moved (pod deleted mid-create) — the event for

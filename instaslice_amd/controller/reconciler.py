@@ -11,8 +11,12 @@ the call stack documented in SURVEY.md §3.1/§3.4. State machine (unchanged):
 Differences by design (not omissions):
   - placement is mode-aware (controller/policy.py) instead of MIG-slot
     first-fit;
-  - all CR mutations go through update_with_retry (conflict-absorbing)
-    instead of a 1 s requeue on conflict;
+  - CR mutations are single-round-trip guarded PATCHes (CAS on
+    resourceVersion for placement, test-ops elsewhere); a failed guard
+    falls back to the classic get-revalidate-update cycle instead of the
+    reference's 1 s requeue-on-conflict (instaslice_controller.go:93);
+  - reconciles read the engine's informer cache (watch-fed) rather than
+    GET/LISTing per reconcile;
   - errors from the placer/store fail the reconcile loudly (the engine
     backs off and retries) instead of being logged and dropped.
 """

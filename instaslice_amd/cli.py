@@ -33,11 +33,11 @@ from instaslice_amd.utils import get_logger
 log = get_logger("cli")
 
 
-def _connect(store_arg: str):
+def _connect(store_arg: str, reconnect: bool = False):
     from instaslice_amd.store.netstore import NetStoreClient
 
     host, _, port = store_arg.rpartition(":")
-    return NetStoreClient(host or "127.0.0.1", int(port))
+    return NetStoreClient(host or "127.0.0.1", int(port), reconnect=reconnect)
 
 
 def _wait_forever():
@@ -78,7 +78,7 @@ def cmd_controller(args) -> int:
     from instaslice_amd.controller.reconciler import PodController
     from instaslice_amd.metrics import get_metrics, serve_http
 
-    store = _connect(args.store)
+    store = _connect(args.store, reconnect=True)
     lease = None
     if args.leader_elect:
         from instaslice_amd.runtime.lease import LeaderLease
@@ -110,7 +110,7 @@ def cmd_daemonset(args) -> int:
     from instaslice_amd.metrics import get_metrics, serve_http
 
     node = args.node_name or os.environ.get("NODE_NAME") or socket.gethostname()
-    store = _connect(args.store)
+    store = _connect(args.store, reconnect=True)
     if args.fake:
         from instaslice_amd.smi.fake import FakeAmdSmi
 

@@ -215,9 +215,21 @@ class _ClientWatch:
 
 
 class NetStoreClient:
-    """Store client with the MemStore interface (duck-typed)."""
+    """Store client with the MemStore interface (duck-typed).
 
-    def __init__(self, host: str, port: int, timeout: float = 30.0) -> None:
+    With reconnect=True (long-running daemons), a lost connection is
+    re-established with exponential backoff and every live watch is
+    re-subscribed WITH REPLAY — the k8s informer-resync analog. Consumers
+    are level-triggered reconcilers, so the duplicate ADDED events a resync
+    produces are harmless; in-flight calls during the outage fail with
+    ConnectionError and the engines' error backoff retries them."""
+
+    def __init__(self, host: str, port: int, timeout: float = 30.0,
+                 reconnect: bool = False,
+                 reconnect_backoff_s: float = 0.2) -> None:
+        self.host, self.port = host, port
+        self._reconnect = reconnect
+        self._reconnect_backoff_s = reconnect_backoff_s
         self._sock = socket.create_connection((host, port), timeout=timeout)
         self._sock.settimeout(None)
         self._sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
@@ -225,6 +237,8 @@ class NetStoreClient:
         self._wlock = threading.Lock()
         self._pending: dict = {}
         self._watches: dict = {}
+        # wid -> (watch, request_kwargs) for reconnect re-subscription
+        self._watch_specs: dict = {}
         # events that arrived before watch() registered their id (the reader
         # thread outruns the caller between response and registration);
         # _watch_reg_lock makes register-vs-deliver atomic
@@ -238,35 +252,110 @@ class NetStoreClient:
                                         name="netstore-reader", daemon=True)
         self._reader.start()
 
-    def _read_loop(self) -> None:
-        try:
-            while True:
-                msg = _read_msg(self._rfile)
-                if msg is None:
-                    break
-                if "watch_id" in msg and "event" in msg:
-                    wid = msg["watch_id"]
-                    et, obj = msg["event"]
-                    with self._watch_reg_lock:
-                        w = self._watches.get(wid)
-                        if w is None:
-                            self._orphan_events.setdefault(wid, []).append((et, obj))
-                    if w is not None and not w._stopped:
-                        w._q.put((et, obj))
-                    continue
-                ev = self._pending.pop(msg.get("id"), None)
-                if ev is not None:
-                    ev[1] = msg
-                    ev[0].set()
-        except (OSError, ValueError):
-            pass
-        finally:
-            self._closed = True
-            for ev in list(self._pending.values()):
-                ev[1] = {"ok": False, "error": {"type": "Error", "msg": "connection closed"}}
+    def _dispatch(self, msg: dict) -> None:
+        if "watch_id" in msg and "event" in msg:
+            wid = msg["watch_id"]
+            et, obj = msg["event"]
+            with self._watch_reg_lock:
+                w = self._watches.get(wid)
+                if w is None:
+                    self._orphan_events.setdefault(wid, []).append((et, obj))
+            if w is not None and not w._stopped:
+                w._q.put((et, obj))
+            return
+        ev = self._pending.pop(msg.get("id"), None)
+        if ev is not None:
+            ev[1] = msg
+            ev[0].set()
+
+    def _fail_pending(self, why: str) -> None:
+        for rid in list(self._pending):
+            ev = self._pending.pop(rid, None)
+            if ev is not None:
+                ev[1] = {"ok": False, "error": {"type": "Error", "msg": why}}
                 ev[0].set()
-            for w in self._watches.values():
-                w.stop()
+
+    def _read_loop(self) -> None:
+        while True:
+            try:
+                while True:
+                    msg = _read_msg(self._rfile)
+                    if msg is None:
+                        break
+                    self._dispatch(msg)
+            except (OSError, ValueError):
+                pass
+            self._fail_pending("connection lost")
+            if self._closed or not self._reconnect:
+                break
+            if not self._reconnect_once():
+                break
+        self._closed = True
+        self._fail_pending("connection closed")
+        for w in self._watches.values():
+            w.stop()
+
+    def _reconnect_once(self) -> bool:
+        """Re-establish the connection and re-subscribe live watches.
+        Runs ON the reader thread (reads its own responses inline)."""
+        try:
+            self._sock.close()
+        except OSError:
+            pass
+        delay = self._reconnect_backoff_s
+        while not self._closed:
+            try:
+                sock = socket.create_connection((self.host, self.port),
+                                                timeout=5.0)
+            except OSError:
+                import time as _time
+
+                _time.sleep(delay)
+                delay = min(delay * 2, 5.0)
+                continue
+            sock.settimeout(None)
+            sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            with self._wlock:
+                self._sock = sock
+                self._rfile = sock.makefile("rb")
+            self._fail_pending("connection lost")  # sent into the dead socket
+            with self._watch_reg_lock:
+                live = [(w, spec) for (w, spec) in self._watch_specs.values()
+                        if not w._stopped]
+                self._watches = {}
+                self._watch_specs = {}
+            ok = True
+            for w, spec in live:
+                with self._idlock:
+                    self._next_id += 1
+                    rid = self._next_id
+                resync = dict(spec, replay=True)
+                try:
+                    with self._wlock:
+                        self._sock.sendall(_pack({"id": rid, "verb": "watch",
+                                                  **resync}))
+                    while True:  # drain events of earlier re-subs inline
+                        msg = _read_msg(self._rfile)
+                        if msg is None:
+                            ok = False
+                            break
+                        if msg.get("id") == rid:
+                            new_wid = msg["result"]["watch_id"]
+                            with self._watch_reg_lock:
+                                self._watches[new_wid] = w
+                                self._watch_specs[new_wid] = (w, spec)
+                            break
+                        self._dispatch(msg)
+                except (OSError, ValueError, KeyError):
+                    ok = False
+                if not ok:
+                    break
+            if ok:
+                self.log.warning(
+                    "reconnected to store %s:%d (%d watches resynced)",
+                    self.host, self.port, len(live))
+                return True
+        return False
 
     def _call(self, verb: str, **kw):
         if self._closed:
@@ -277,8 +366,12 @@ class NetStoreClient:
         ev = [threading.Event(), None]
         self._pending[rid] = ev
         req = _pack({"id": rid, "verb": verb, **kw})
-        with self._wlock:
-            self._sock.sendall(req)
+        try:
+            with self._wlock:
+                self._sock.sendall(req)
+        except OSError as e:
+            self._pending.pop(rid, None)
+            raise ConnectionError(f"netstore send failed: {e}") from e
         if not ev[0].wait(timeout=60.0):
             self._pending.pop(rid, None)
             raise TimeoutError(f"netstore call {verb} timed out")
@@ -286,6 +379,8 @@ class NetStoreClient:
         if msg["ok"]:
             return msg.get("result")
         err = msg["error"]
+        if err["type"] == "Error" and "connection" in err["msg"]:
+            raise ConnectionError(err["msg"])
         raise _EXC_BY_NAME.get(err["type"], RuntimeError)(err["msg"])
 
     # -- MemStore interface -------------------------------------------------
@@ -317,10 +412,12 @@ class NetStoreClient:
               filters: Optional[List[dict]] = None):
         w = _ClientWatch()
         w.kind = kind
+        spec = {"kind": kind, "replay": replay, "filters": filters}
         res = self._call("watch", kind=kind, replay=replay, filters=filters)
         wid = res["watch_id"]
         with self._watch_reg_lock:
             self._watches[wid] = w
+            self._watch_specs[wid] = (w, spec)
             orphans = self._orphan_events.pop(wid, [])
         # deliver events that raced ahead of this registration, in order
         for ev in orphans:

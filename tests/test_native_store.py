@@ -284,3 +284,36 @@ def test_native_full_pod_lifecycle_stack():
         for c in (cstore, astore, bench):
             c.close()
         server.stop()
+
+
+@needs_stored
+def test_client_reconnects_after_store_restart():
+    """Kill the store daemon mid-session and restart it on the same port:
+    a reconnect=True client must resume — calls work again and its watches
+    are re-subscribed with replay (informer resync)."""
+    server = NativeStoreServer().start()
+    port = server.port
+    c = NetStoreClient("127.0.0.1", port, reconnect=True)
+    try:
+        c.create(_obj("survivor"))
+        w = c.watch("Thing", replay=False)
+        server.stop()  # daemon killed; state is gone (no persistence)
+        time.sleep(0.3)
+        server = NativeStoreServer(port=port).start()
+        # calls recover (engine error-backoff analog: retry until live)
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            try:
+                c.create(_obj("reborn"))
+                break
+            except (ConnectionError, TimeoutError, RuntimeError):
+                time.sleep(0.2)
+        else:
+            pytest.fail("client never recovered")
+        # the re-subscribed watch sees post-restart events
+        ev = w.next(timeout=5)
+        assert ev is not None and ev[1]["metadata"]["name"] == "reborn"
+        w.stop()
+    finally:
+        c.close()
+        server.stop()

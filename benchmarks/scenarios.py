@@ -143,14 +143,18 @@ def scenario_mixed100(args) -> dict:
             outcome = c.wait_pod_outcome(name, timeout=10.0)
             while outcome == "unschedulable":
                 # out of capacity: drain oldest pods one at a time until the
-                # request fits (mixed modes may need a whole GPU to go idle)
+                # request fits (mixed modes may need a whole GPU to go idle).
+                # Wait for the CAPACITY to free (allocation drained), then
+                # give the event-driven re-place a short window — the
+                # controller reacts to the CR event in single-digit ms.
                 if not live:
                     raise RuntimeError(f"{name} unplaceable on empty node")
                 old = live.pop(0)
                 c.delete_pod(old)
                 c.wait_pod_gone(old)
+                c.wait_pod_unallocated(old)
                 try:
-                    c.wait_pod_scheduled(name, timeout=1.0)
+                    c.wait_pod_scheduled(name, timeout=0.25)
                     outcome = "scheduled"
                 except TimeoutError:
                     continue

@@ -119,6 +119,20 @@ class Cluster:
             time.sleep(0.002)
         raise TimeoutError(f"pod {namespace}/{name} still present after {timeout}s")
 
+    def wait_pod_unallocated(self, pod_name: str, timeout: float = 10.0) -> None:
+        """Block until no node CR carries an allocation for `pod_name`
+        (capacity actually freed — the moment waiting pods can re-place)."""
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            if not any(
+                a.get("podName") == pod_name
+                for cr in self.store.list("Instaslice")
+                for a in (cr["spec"].get("allocations") or {}).values()
+            ):
+                return
+            time.sleep(0.005)
+        raise TimeoutError(f"allocation for {pod_name} never drained")
+
     def wait_allocations_empty(self, node: str, timeout: float = 10.0) -> None:
         deadline = time.monotonic() + timeout
         while time.monotonic() < deadline:

@@ -83,8 +83,14 @@ def cmd_controller(args) -> int:
     if args.leader_elect:
         from instaslice_amd.runtime.lease import LeaderLease
 
-        lease = LeaderLease(store, "instaslice-controller").start()
-        log.info("waiting for leadership...")
+        # one lease PER SHARD: replicas of the same shard compete, distinct
+        # shards run concurrently (a single shared lease would serialize the
+        # whole sharded deployment down to one controller)
+        lease_name = ("instaslice-controller"
+                      if args.shard_count <= 1
+                      else f"instaslice-controller-shard-{args.shard_index}")
+        lease = LeaderLease(store, lease_name).start()
+        log.info("waiting for leadership of %s...", lease_name)
         lease.wait_leader()
     controller = PodController(store, policy=args.policy,
                                teardown_grace_s=args.grace,

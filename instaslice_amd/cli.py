@@ -54,14 +54,12 @@ def cmd_store(args) -> int:
     from instaslice_amd.store.netstore import StoreServer
 
     backing = None
-    use_native = (args.native or (args.native is None and not args.data
-                                  and stored_available()))
+    use_native = (args.native or (args.native is None and stored_available()))
     if use_native:
-        if args.data:
-            raise SystemExit("--native store has no persistence; drop --data")
-        server = NativeStoreServer(port=args.port).start()
-        log.info("native store (instaslice-stored) serving on %s:%d",
-                 server.host, server.port)
+        server = NativeStoreServer(port=args.port,
+                                   persist_path=args.data).start()
+        log.info("native store (instaslice-stored) serving on %s:%d (data=%s)",
+                 server.host, server.port, args.data or "in-memory")
     else:
         backing = MemStore(persist_path=args.data) if args.data else None
         server = StoreServer(store=backing, port=args.port).start()
@@ -217,8 +215,9 @@ def main(argv=None) -> int:
     p = sub.add_parser("store", help="run the state store server")
     p.add_argument("--port", type=int, default=7080)
     p.add_argument("--data", default=None,
-                   help="persist state to this JSON file (checkpoint/resume; "
-                        "forces the Python server)")
+                   help="persist state to this file (checkpoint/resume; "
+                        "JSON under the Python server, msgpack under the "
+                        "native daemon)")
     p.add_argument("--native", action="store_true", default=None,
                    help="serve with the C++ instaslice-stored daemon "
                         "(default: auto when built and --data is unset)")

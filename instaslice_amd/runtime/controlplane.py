@@ -74,10 +74,9 @@ def run_control_plane(
     (one shard saturates near ~400 pods/s).
 
     native_store (None = auto): run the C++ store daemon
-    (instaslice-stored) instead of the Python server. Auto picks it for
-    sharded mode when the binary is built and no persistence is requested —
-    the Python server's GIL-bound wire handling is the next ceiling once
-    controllers shard."""
+    (instaslice-stored) instead of the Python server whenever the binary is
+    built — the Python server's GIL-bound wire handling caps throughput.
+    Both support persist_path (checkpoint/resume)."""
     # short GIL switch interval: the reconcile path is wakeup-latency bound
     sys.setswitchinterval(0.001)
     from instaslice_amd.controller.reconciler import PodController
@@ -89,11 +88,12 @@ def run_control_plane(
         # measured: the daemon beats the Python server even for a single
         # rank WITH the controller paying TCP round-trips (p50 2.9 vs 4.9 ms
         # on the dev box) — the Python server's wire handling costs more
-        # than the extra hops. Python remains for persistence mode.
-        native_store = persist_path is None and stored_available()
+        # than the extra hops. Both back persistence (JSON for Python,
+        # msgpack snapshots for the daemon).
+        native_store = stored_available()
     store = None
     if native_store:
-        server = NativeStoreServer(port=port).start()
+        server = NativeStoreServer(port=port, persist_path=persist_path).start()
     else:
         store = MemStore(persist_path=persist_path)
         server = StoreServer(store=store, port=port).start()

@@ -38,6 +38,7 @@
 #include <chrono>
 #include <condition_variable>
 #include <cstdio>
+#include <csignal>
 #include <cstring>
 #include <deque>
 #include <map>
@@ -199,6 +200,7 @@ class Store {
       out = *stored;
     }
     for (Conn* c : wake) c->out_cv.notify_one();
+    mark_dirty();
     return out;
   }
 
@@ -255,6 +257,7 @@ class Store {
     out = commit_locked(k, std::move(obj), wake);
     }
     for (Conn* c : wake) c->out_cv.notify_one();
+    mark_dirty();
     return out;
   }
 
@@ -291,6 +294,7 @@ class Store {
     }
     }
     for (Conn* c : wake) c->out_cv.notify_one();
+    mark_dirty();
   }
 
   Value patch(const std::string& kind, const std::string& ns,
@@ -312,6 +316,7 @@ class Store {
       out = commit_locked(k, std::move(obj), wake);
     }
     for (Conn* c : wake) c->out_cv.notify_one();
+    mark_dirty();
     return out;
   }
 
@@ -524,6 +529,68 @@ class Store {
   std::map<std::string, ObjPtr> objects_;
   uint64_t rv_ = 0;
   std::vector<std::shared_ptr<WatchSub>> watches_;
+
+  // -- persistence (checkpoint/resume; msgpack snapshot, write-behind) ----
+  // Python MemStore snapshots JSON; the daemon snapshots the same
+  // {"rv": N, "objects": [...]} shape in msgpack. Debounced: bursts of
+  // mutations coalesce into one write, and SIGTERM flushes synchronously.
+  std::string persist_path_;
+  std::atomic<bool> persist_dirty_{false};
+
+ public:
+  void set_persist_path(std::string path) { persist_path_ = std::move(path); }
+
+  void mark_dirty() {
+    if (!persist_path_.empty()) persist_dirty_.store(true);
+  }
+
+  bool load_snapshot() {
+    if (persist_path_.empty()) return false;
+    FILE* f = std::fopen(persist_path_.c_str(), "rb");
+    if (!f) return false;
+    std::string buf;
+    char tmp[1 << 16];
+    size_t n;
+    while ((n = std::fread(tmp, 1, sizeof(tmp), f)) > 0) buf.append(tmp, n);
+    std::fclose(f);
+    try {
+      Value snap = unpack(buf);
+      const Value* rv = snap.find("rv");
+      const Value* objs = snap.find("objects");
+      std::lock_guard<std::mutex> g(mu_);
+      if (rv && rv->t == Value::T::Int) rv_ = static_cast<uint64_t>(rv->i);
+      if (objs && objs->is_arr())
+        for (const auto& o : *objs->a)
+          objects_[obj_key(o)] = std::make_shared<const Value>(o);
+      return true;
+    } catch (const std::exception& e) {
+      std::fprintf(stderr, "snapshot load failed: %s\n", e.what());
+      return false;
+    }
+  }
+
+  void flush_snapshot() {
+    if (persist_path_.empty()) return;
+    Value snap = Value::map();
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      snap.setkey("rv", Value::integer(static_cast<int64_t>(rv_)));
+      Value arr = Value::arr();
+      arr.a->reserve(objects_.size());
+      for (const auto& kv : objects_) arr.a->push_back(*kv.second);
+      snap.setkey("objects", std::move(arr));
+    }
+    std::string payload;  // packed OUTSIDE the lock (objects are immutable)
+    pack(snap, payload);
+    std::string tmp_path = persist_path_ + ".tmp";
+    FILE* f = std::fopen(tmp_path.c_str(), "wb");
+    if (!f) return;
+    std::fwrite(payload.data(), 1, payload.size(), f);
+    std::fclose(f);
+    std::rename(tmp_path.c_str(), persist_path_.c_str());
+  }
+
+  bool take_dirty() { return persist_dirty_.exchange(false); }
 };
 
 // ---- per-connection handling ----------------------------------------------
@@ -733,10 +800,15 @@ void reader_loop(Store& store, std::shared_ptr<Conn> conn) {
 
 }  // namespace stored
 
+static std::atomic<bool> g_stop{false};
+
+static void on_signal(int) { g_stop.store(true); }
+
 int main(int argc, char** argv) {
   using namespace stored;
   int port = 0;
   if (argc > 1) port = std::atoi(argv[1]);
+  const char* snapshot = (argc > 2) ? argv[2] : nullptr;
 
   int srv = ::socket(AF_INET, SOCK_STREAM, 0);
   if (srv < 0) { perror("socket"); return 1; }
@@ -754,10 +826,32 @@ int main(int argc, char** argv) {
   ::getsockname(srv, reinterpret_cast<sockaddr*>(&addr), &alen);
   if (::listen(srv, 64) != 0) { perror("listen"); return 1; }
   // parent (store/native.py) parses this line for the chosen port
+  Store store;
+  if (snapshot) {
+    store.set_persist_path(snapshot);
+    if (store.load_snapshot())
+      std::fprintf(stderr, "resumed from %s\n", snapshot);
+    std::signal(SIGTERM, on_signal);
+    std::signal(SIGINT, on_signal);
+    // write-behind snapshot thread (200 ms debounce) + SIGTERM flush
+    std::thread([&store] {
+      for (;;) {
+        std::this_thread::sleep_for(std::chrono::milliseconds(100));
+        if (g_stop.load()) {
+          store.flush_snapshot();
+          std::_Exit(0);
+        }
+        if (store.take_dirty()) {
+          std::this_thread::sleep_for(std::chrono::milliseconds(100));
+          store.take_dirty();
+          store.flush_snapshot();
+        }
+      }
+    }).detach();
+  }
   std::printf("LISTENING %d\n", ntohs(addr.sin_port));
   std::fflush(stdout);
 
-  Store store;
   for (;;) {
     int fd = ::accept(srv, nullptr, nullptr);
     if (fd < 0) {

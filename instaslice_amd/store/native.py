@@ -27,17 +27,22 @@ class NativeStoreServer:
     Mirrors StoreServer's start/stop/port surface (no .store attribute —
     state lives in the daemon; use a NetStoreClient)."""
 
-    def __init__(self, port: int = 0, binary: Optional[str] = None) -> None:
+    def __init__(self, port: int = 0, binary: Optional[str] = None,
+                 persist_path: Optional[str] = None) -> None:
         self._binary = binary or STORED_BIN
         self._want_port = port
+        self._persist_path = persist_path
         self._proc: Optional[subprocess.Popen] = None
         self.port: Optional[int] = None
         self.host = "127.0.0.1"
         self.log = get_logger("netstore.native")
 
     def start(self) -> "NativeStoreServer":
+        cmd = [self._binary, str(self._want_port)]
+        if self._persist_path:
+            cmd.append(self._persist_path)
         self._proc = subprocess.Popen(
-            [self._binary, str(self._want_port)],
+            cmd,
             stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True,
         )
         line = self._proc.stdout.readline().strip()

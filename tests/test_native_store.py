@@ -317,3 +317,33 @@ def test_client_reconnects_after_store_restart():
     finally:
         c.close()
         server.stop()
+
+
+@needs_stored
+def test_native_persistence_checkpoint_resume(tmp_path):
+    """SIGTERM flushes a msgpack snapshot; a fresh daemon resumes with the
+    same objects AND resourceVersion monotonicity (the Python MemStore's
+    persist_path contract, store/memstore.py)."""
+    snap = str(tmp_path / "state.msgpack")
+    server = NativeStoreServer(persist_path=snap).start()
+    c = NetStoreClient("127.0.0.1", server.port)
+    c.create(_obj("d1", spec={"v": 1}))
+    c.create(_obj("d2"))
+    c.patch("Thing", "d1", "", [{"op": "set", "path": ["spec", "v"], "value": 2}])
+    rv_before = c.get("Thing", "d1")["metadata"]["resourceVersion"]
+    c.close()
+    server.stop()  # SIGTERM -> flush
+
+    server = NativeStoreServer(persist_path=snap).start()
+    c = NetStoreClient("127.0.0.1", server.port)
+    try:
+        got = c.get("Thing", "d1")
+        assert got["spec"]["v"] == 2
+        assert got["metadata"]["resourceVersion"] == rv_before
+        assert len(c.list("Thing")) == 2
+        # new writes continue the rv sequence past the resumed point
+        c.patch("Thing", "d2", "", [{"op": "set", "path": ["x"], "value": 1}])
+        assert int(c.get("Thing", "d2")["metadata"]["resourceVersion"]) > int(rv_before)
+    finally:
+        c.close()
+        server.stop()

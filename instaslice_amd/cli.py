@@ -72,6 +72,42 @@ def cmd_store(args) -> int:
     return 0
 
 
+def cmd_controlplane(args) -> int:
+    """Store + sharded controllers in one command (runtime/controlplane):
+    the single-node quick-start and bench topology."""
+    import multiprocessing as mp
+
+    from instaslice_amd.runtime.controlplane import run_control_plane
+
+    ctx = mp.get_context("spawn")
+    parent_conn, child_conn = ctx.Pipe()
+    proc = ctx.Process(
+        target=run_control_plane, args=(child_conn, args.policy),
+        kwargs={
+            "teardown_grace_s": args.grace,
+            "port": args.port,
+            "persist_path": args.data,
+            "controller_shards": args.shards,
+        },
+        name="control-plane",
+    )
+    proc.start()
+    port = parent_conn.recv()
+    log.info("control plane on 127.0.0.1:%d (%d controller shard(s))",
+             port, max(1, args.shards))
+    try:
+        _wait_forever()
+    finally:
+        try:
+            parent_conn.send("stop")
+        except (BrokenPipeError, OSError):
+            pass
+        proc.join(timeout=5.0)
+        if proc.is_alive():
+            proc.terminate()
+    return 0
+
+
 def cmd_controller(args) -> int:
     from instaslice_amd.controller.reconciler import PodController
     from instaslice_amd.metrics import get_metrics, serve_http
@@ -224,6 +260,17 @@ def main(argv=None) -> int:
     p.add_argument("--no-native", dest="native", action="store_false",
                    help="force the Python store server")
     p.set_defaults(fn=cmd_store)
+
+    p = sub.add_parser("controlplane",
+                       help="store + sharded controllers in one command")
+    p.add_argument("--port", type=int, default=7080)
+    p.add_argument("--policy", default="packed-fit")
+    p.add_argument("--grace", type=float, default=30.0)
+    p.add_argument("--shards", type=int, default=1,
+                   help="controller shard processes")
+    p.add_argument("--data", default=None,
+                   help="persist state (checkpoint/resume)")
+    p.set_defaults(fn=cmd_controlplane)
 
     p = sub.add_parser("controller", help="run the cluster controller")
     p.add_argument("--store", default="127.0.0.1:7080")

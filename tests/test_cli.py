@@ -144,3 +144,41 @@ def test_cli_native_store_sharded_controllers():
             except subprocess.TimeoutExpired:
                 p.kill()
         server.stop()
+
+
+def test_cli_controlplane_one_command():
+    """`controlplane` = store + controllers in one process tree."""
+    import re
+
+    p = subprocess.Popen(
+        [sys.executable, "-m", "instaslice_amd", "controlplane", "--port", "0",
+         "--grace", "0", "--shards", "1"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    agent = None
+    try:
+        port = None
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            line = p.stdout.readline()
+            m = re.search(r"control plane on 127\.0\.0\.1:(\d+)", line or "")
+            if m:
+                port = int(m.group(1))
+                break
+        assert port, "control plane never reported its port"
+        agent = subprocess.Popen(
+            [sys.executable, "-m", "instaslice_amd", "daemonset", "--store",
+             f"127.0.0.1:{port}", "--node-name", "cp-node", "--fake", "1",
+             "--metrics-port", "0"],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "submit", "--store",
+             f"127.0.0.1:{port}", "--name", "cppod", "--profile", "spx-8x288",
+             "--wait", "--timeout", "30"],
+            capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stdout + out.stderr
+    finally:
+        if agent:
+            agent.terminate()
+            agent.wait(timeout=5)
+        p.terminate()
+        p.wait(timeout=10)

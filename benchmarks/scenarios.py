@@ -34,7 +34,10 @@ from instaslice_amd.smi.fake import FakeAmdSmi  # noqa: E402
 
 def _mk_cluster(num_gpus: int = 8, nodes: int = 1, policy: str = "packed-fit",
                 compute_latency_s: float = 0.0) -> Cluster:
-    c = Cluster(teardown_grace_s=0.0, policy=policy)
+    from instaslice_amd.store.native import stored_available
+
+    backend = "native" if stored_available() else "mem"
+    c = Cluster(teardown_grace_s=0.0, policy=policy, backend=backend)
     for n in range(nodes):
         c.add_node(
             f"node-{n}",

@@ -26,10 +26,32 @@ class Cluster:
         policy: str = "packed-fit",
         teardown_grace_s: float = 0.0,
         reset_mode_on_empty: bool = False,
+        backend: str = "mem",
     ) -> None:
-        self.store = store or MemStore()
+        # backend="native" serves state from the C++ store daemon and gives
+        # each component its own TCP client — the production data plane
+        # inside the single-process harness (requires instaslice-stored;
+        # raises if it isn't built)
+        self._native_server = None
+        self._clients: List = []
+        if store is None and backend == "native":
+            from instaslice_amd.store.native import NativeStoreServer
+            from instaslice_amd.store.netstore import NetStoreClient
+
+            self._native_server = NativeStoreServer().start()
+
+            def _client():
+                c = NetStoreClient("127.0.0.1", self._native_server.port)
+                self._clients.append(c)
+                return c
+
+            self._mk_store = _client
+            self.store = _client()
+        else:
+            self.store = store or MemStore()
+            self._mk_store = lambda: self.store
         self.controller = PodController(
-            self.store, policy=policy, teardown_grace_s=teardown_grace_s
+            self._mk_store(), policy=policy, teardown_grace_s=teardown_grace_s
         )
         self.agents: Dict[str, NodeAgent] = {}
         self.reset_mode_on_empty = reset_mode_on_empty
@@ -38,7 +60,8 @@ class Cluster:
 
     def add_node(self, node_name: str, smi: AmdSmi) -> NodeAgent:
         agent = NodeAgent(
-            self.store, smi, node_name, reset_mode_on_empty=self.reset_mode_on_empty
+            self._mk_store(), smi, node_name,
+            reset_mode_on_empty=self.reset_mode_on_empty,
         )
         self.agents[node_name] = agent
         if self._started:
@@ -56,6 +79,10 @@ class Cluster:
         self.controller.stop()
         for agent in self.agents.values():
             agent.stop()
+        for c in self._clients:
+            c.close()
+        if self._native_server is not None:
+            self._native_server.stop()
 
     # -- workload helpers --------------------------------------------------
 

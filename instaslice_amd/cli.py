@@ -1,10 +1,13 @@
 """Operational entry points — the analog of the reference's two binaries
 (cmd/controller/main.go, cmd/daemonset/main.go) plus operator conveniences.
 
-    python -m instaslice_amd store      [--port 7080]
+    python -m instaslice_amd controlplane [--port 7080] [--shards K]
+                                        [--data STATE] [--policy packed-fit]
+    python -m instaslice_amd store      [--port 7080] [--native] [--data STATE]
     python -m instaslice_amd controller --store HOST:PORT [--policy packed-fit]
                                         [--metrics-port 8080] [--grace 30]
                                         [--leader-elect]
+                                        [--shard-index I --shard-count K]
     python -m instaslice_amd daemonset  --store HOST:PORT [--node-name NAME]
                                         [--fake N] [--metrics-port 8084]
                                         [--reset-on-empty]

@@ -12,7 +12,7 @@ import time
 from typing import Dict, List, Optional
 
 from instaslice_amd.agent.daemonset import NodeAgent
-from instaslice_amd.api.types import AllocationStatus, new_pod
+from instaslice_amd.api.types import new_pod
 from instaslice_amd.controller.reconciler import INSTASLICE_NS, PodController
 from instaslice_amd.smi.base import AmdSmi
 from instaslice_amd.store.memstore import MemStore, NotFound

@@ -15,7 +15,7 @@ processor-count identity mapping (1 processor == 1 GPU).
 from __future__ import annotations
 
 import threading
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from instaslice_amd.smi.base import (
     AmdSmi,

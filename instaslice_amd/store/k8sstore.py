@@ -20,7 +20,7 @@ from __future__ import annotations
 
 import queue
 import threading
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Optional
 
 from instaslice_amd import API_GROUP, API_VERSION
 from instaslice_amd.store.memstore import AlreadyExists, Conflict, NotFound

@@ -70,8 +70,11 @@ class NodeAgent:
         # cached enumeration: gpu uuid -> PhysicalGpu (refreshed only after a
         # mode change of that GPU)
         self._gpus: Dict[str, PhysicalGpu] = {}
-        # observability feed (metrics module scrapes this)
-        self.reconfigure_events: List[dict] = []
+        # observability feed (metrics module scrapes this; bounded so a
+        # long-running agent cannot grow without limit)
+        from collections import deque
+
+        self.reconfigure_events: deque = deque(maxlen=4096)
         # consecutive hard mode-set failures per GPU (3 strikes -> fail alloc)
         self._mode_set_failures: Dict[str, int] = {}
         self.prepare_failures = 0  # observability counter

@@ -83,8 +83,11 @@ class PodController:
         # — externally crafted — are treated as healthy.)
         self.node_stale_after_s = node_stale_after_s
         self.log = get_logger("controller")
-        # latency observability: pod_uid -> submit time, and measured p50 feed
-        self.alloc_latency_s: List[float] = []
+        # latency observability: pod_uid -> submit time, and measured p50
+        # feed (bounded: long-running controllers must not grow unboundedly)
+        from collections import deque
+
+        self.alloc_latency_s: deque = deque(maxlen=65536)
         self._pending_since: dict = {}
         # pods currently marked unschedulable: re-reconciled on ANY node-state
         # change so freed capacity is picked up event-driven, not by polling
@@ -360,6 +363,7 @@ class PodController:
         # teardown path (reference: instaslice_controller.go:99-142)
         if md.get("deletionTimestamp"):
             self._unschedulable_keys.discard((namespace, name))
+            self._pending_since.pop(uid, None)  # deleted before ungating
             gated = pod_is_gated(pod)
             if not gated:
                 elapsed = time.time() - float(md["deletionTimestamp"])

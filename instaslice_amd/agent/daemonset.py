@@ -229,6 +229,20 @@ class NodeAgent:
                 self.node_name, f"{compute}/{memory}", event["set_wall_ms"] / 1000.0
             )
             g = self._gpus[gpu_uuid]
+            # device-plugin nudge (reference: label-toggle reload trick,
+            # instaslice_daemonset.go:474-497). The AMD plugin re-enumerates
+            # KFD on its health interval so no reload is required; this
+            # label is the operator-visible signal (and a hook for plugins
+            # that DO watch labels): bumped after every realized mode change
+            try:
+                self.store.patch("Node", self.node_name, "", [
+                    {"op": "set",
+                     "path": ["metadata", "labels",
+                              "org.instaslice/last-reconfigure"],
+                     "value": str(int(time.time()))},
+                ], quiet=True)
+            except Exception as e:  # noqa: BLE001 - label nudge is best-effort
+                self.log.warning("device-plugin nudge failed: %s", e)
             self.log.info(
                 "gpu %s reconfigured %s -> %s/%s in %.1f ms (%d partitions)",
                 gpu_uuid[:8], event["from"], g.compute_mode, g.memory_mode,

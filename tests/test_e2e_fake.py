@@ -387,3 +387,23 @@ def test_sharded_controllers_partition_pod_ownership():
         for c in shards:
             c.stop()
         agent.stop()
+
+
+def test_reconfigure_bumps_device_plugin_label():
+    """Every realized mode change stamps org.instaslice/last-reconfigure on
+    the Node — the AMD-device-plugin re-advertisement signal (reference
+    label-toggle analog, instaslice_daemonset.go:474-497)."""
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        node = c.store.get("Node", "node-0", "")
+        assert "org.instaslice/last-reconfigure" not in (
+            node["metadata"].get("labels") or {})
+        c.submit_pod("dp0", "cpx-1x36")  # SPX -> CPX flip
+        c.wait_pod_scheduled("dp0")
+        node = c.store.get("Node", "node-0", "")
+        assert (node["metadata"].get("labels") or {}).get(
+            "org.instaslice/last-reconfigure")
+    finally:
+        c.stop()

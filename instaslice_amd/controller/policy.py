@@ -127,27 +127,43 @@ def _target_memory_mode(profile: PartitionProfile, current: MemoryMode) -> Tuple
     ), True
 
 
+Score = Tuple  # lexicographic, higher wins; COMPARABLE ACROSS NODES
+
 class AllocationPolicy:
     """Strategy interface (reference: instaslice_controller.go:48-50).
 
     `prefer_gpus` carries gang-placement affinity: GPUs already hosting the
     pod's group. Same-GPU XCD co-location is free bandwidth (intra-die,
     no xGMI hop — SURVEY.md §5), so preferred same-mode fits win over
-    everything else in every policy."""
+    everything else in every policy.
+
+    `place_scored` returns (score, placement) with scores comparable across
+    nodes: the controller collects one candidate per node and takes the
+    global argmax, so packing/spreading works cluster-wide, not just within
+    the first node that fits (the reference is first-fit across nodes,
+    findDeviceForASlice, instaslice_controller.go:240-262)."""
 
     name = "base"
+    node_order_first_fit = False  # True: controller takes the first node hit
+
+    def place_scored(self, profile: PartitionProfile, views: List[GpuView],
+                     prefer_gpus: frozenset = frozenset()
+                     ) -> Optional[Tuple[Score, Placement]]:
+        raise NotImplementedError
 
     def place(self, profile: PartitionProfile, views: List[GpuView],
               prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
-        raise NotImplementedError
+        sp = self.place_scored(profile, views, prefer_gpus)
+        return sp[1] if sp else None
 
     def _preferred(self, profile: PartitionProfile, views: List[GpuView],
-                   prefer_gpus: frozenset) -> Optional[Placement]:
+                   prefer_gpus: frozenset
+                   ) -> Optional[Tuple[Score, Placement]]:
         for v in views:
             if v.uuid in prefer_gpus and v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
                 if p:
-                    return p
+                    return ((3, 0, 0), p)  # gang affinity: top tier
         return None
 
 
@@ -179,23 +195,25 @@ class FirstFitPolicy(AllocationPolicy):
     the right mode, else the first idle GPU (planning a mode change)."""
 
     name = "first-fit"
+    node_order_first_fit = True
 
-    def place(self, profile: PartitionProfile, views: List[GpuView],
-              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
+    def place_scored(self, profile: PartitionProfile, views: List[GpuView],
+                     prefer_gpus: frozenset = frozenset()
+                     ) -> Optional[Tuple[Score, Placement]]:
         if prefer_gpus:
-            p = self._preferred(profile, views, prefer_gpus)
-            if p:
-                return p
+            sp = self._preferred(profile, views, prefer_gpus)
+            if sp:
+                return sp
         for v in views:
             if v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
                 if p:
-                    return p
+                    return ((2, 0, 0), p)
         for v in views:
             if v.compute_mode is not profile.compute:
                 p = _place_on(v, profile, needs_change=True)
                 if p:
-                    return p
+                    return ((1, 0, 0), p)
         return None
 
 
@@ -216,12 +234,13 @@ class PackedFitPolicy(AllocationPolicy):
 
     name = "packed-fit"
 
-    def place(self, profile: PartitionProfile, views: List[GpuView],
-              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
+    def place_scored(self, profile: PartitionProfile, views: List[GpuView],
+                     prefer_gpus: frozenset = frozenset()
+                     ) -> Optional[Tuple[Score, Placement]]:
         if prefer_gpus:
-            p = self._preferred(profile, views, prefer_gpus)
-            if p:
-                return p
+            sp = self._preferred(profile, views, prefer_gpus)
+            if sp:
+                return sp
         best: Optional[Tuple[Tuple, Placement]] = None
         for v in views:
             if v.compute_mode is profile.compute:
@@ -238,7 +257,7 @@ class PackedFitPolicy(AllocationPolicy):
                 score = (1, 1 if mem_ok else 0, -v.index)
             if best is None or score > best[0]:
                 best = (score, p)
-        return best[1] if best else None
+        return best
 
 
 class SpreadFitPolicy(AllocationPolicy):
@@ -250,12 +269,13 @@ class SpreadFitPolicy(AllocationPolicy):
 
     name = "spread-fit"
 
-    def place(self, profile: PartitionProfile, views: List[GpuView],
-              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
+    def place_scored(self, profile: PartitionProfile, views: List[GpuView],
+                     prefer_gpus: frozenset = frozenset()
+                     ) -> Optional[Tuple[Score, Placement]]:
         if prefer_gpus:
-            p = self._preferred(profile, views, prefer_gpus)
-            if p:
-                return p
+            sp = self._preferred(profile, views, prefer_gpus)
+            if sp:
+                return sp
         best: Optional[Tuple[Tuple, Placement]] = None
         for v in views:
             # occupancy is the primary key: an idle GPU (even one needing a
@@ -274,7 +294,7 @@ class SpreadFitPolicy(AllocationPolicy):
                 score = (0, 0, (1 if mem_ok else 0) - v.index)
             if best is None or score > best[0]:
                 best = (score, p)
-        return best[1] if best else None
+        return best
 
 
 POLICIES: Dict[str, AllocationPolicy] = {

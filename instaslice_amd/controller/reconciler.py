@@ -181,11 +181,16 @@ class PodController:
 
     def _try_place(self, pod: dict, profile_name: str, crs: List[dict]) -> Optional[Result]:
         """Scan nodes, place, persist the allocation (status=creating).
-        Reference: findDeviceForASlice loop (instaslice_controller.go:192-222)."""
+        Reference: findDeviceForASlice loop (instaslice_controller.go:192-222)
+        — but where the reference takes the FIRST node that fits, scored
+        policies here collect one candidate per node and commit the global
+        argmax, so packing/spreading is cluster-wide. FirstFit keeps the
+        reference's take-the-first semantics."""
         md = pod["metadata"]
         node_selector = pod.get("spec", {}).get("nodeSelector") or {}
         want_node = node_selector.get("kubernetes.io/hostname")
         group = (md.get("annotations") or {}).get("org.instaslice/group", "")
+        candidates = []  # (score, cr, placement, profile)
         for cr in crs:
             node_name = cr["metadata"]["name"]
             if want_node and node_name != want_node:
@@ -203,9 +208,16 @@ class PodController:
                 for a in (cr["spec"].get("allocations") or {}).values()
                 if group and a.get("group") == group
             )
-            placement = self.policy.place(profile, views, prefer_gpus=prefer)
-            if placement is None:
+            sp = self.policy.place_scored(profile, views, prefer_gpus=prefer)
+            if sp is None:
                 continue
+            candidates.append((sp[0], cr, sp[1], profile))
+            if self.policy.node_order_first_fit:
+                break
+        # best score first; on a lost CAS race, fall through to the next node
+        candidates.sort(key=lambda c: c[0], reverse=True)
+        for _, cr, placement, profile in candidates:
+            node_name = cr["metadata"]["name"]
             alloc = AllocationDetails(
                 profile=profile_name,
                 gpu_uuid=placement.gpu_uuid,
@@ -221,7 +233,6 @@ class PodController:
                 memory_mode=placement.memory_mode,
                 group=group,
             )
-
             updated = self._write_allocation(cr, node_name, alloc)
             if updated and md["uid"] in updated["spec"]["allocations"]:
                 self.log.debug(

@@ -26,7 +26,7 @@ ops_strategy = st.lists(
 )
 
 
-def _settle(c: Cluster, submitted, deleted, timeout=20.0):
+def _settle(c: Cluster, submitted, deleted, timeout=45.0):
     """Wait until every live pod is scheduled or marked unschedulable and
     every deleted pod is fully drained."""
     deadline = time.monotonic() + timeout
@@ -60,7 +60,10 @@ def _settle(c: Cluster, submitted, deleted, timeout=20.0):
             if not (names_alloc & set(deleted)):
                 return
         time.sleep(0.02)
-    raise TimeoutError("cluster never settled")
+    pods = {p["metadata"]["name"]: (bool(p["spec"].get("schedulingGates")),
+                                    p["metadata"].get("annotations"))
+            for p in c.store.list("Pod")}
+    raise TimeoutError(f"cluster never settled; pods={pods}")
 
 
 def _check_invariants(c: Cluster):
@@ -105,7 +108,7 @@ def _check_invariants(c: Cluster):
 
 
 @pytest.mark.parametrize("policy", ["packed-fit", "first-fit", "spread-fit"])
-@settings(max_examples=12, deadline=None,
+@settings(max_examples=10, deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(ops=ops_strategy)
 def test_random_lifecycle_invariants(policy, ops):

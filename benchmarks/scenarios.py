@@ -277,8 +277,48 @@ def scenario_flatlat(args) -> dict:
     return out
 
 
+def scenario_manynode(args) -> dict:
+    """Cluster-scale placement: 32 nodes x 8 GPUs (256 GPUs, 2048 CPX
+    slots), 400 mixed-profile pods with no node pins. Stresses the
+    controller's per-admission scan over all node CRs and the cross-node
+    argmax; reports placement balance across nodes."""
+    n_nodes, n_pods = 32, args.pods or 400
+    c = _mk_cluster(num_gpus=8, nodes=n_nodes, policy=args.policy)
+    rng = random.Random(7)
+    lat = []
+    try:
+        t0 = time.perf_counter()
+        for i in range(n_pods):
+            prof = rng.choice(["cpx-1x36", "qpx-2x72"])
+            t1 = time.perf_counter()
+            c.submit_pod(f"mn-{i}", prof)
+            c.wait_pod_scheduled(f"mn-{i}", timeout=30.0)
+            lat.append((time.perf_counter() - t1) * 1000)
+        wall = time.perf_counter() - t0
+        per_node = {}
+        for cr in c.store.list("Instaslice"):
+            n = len(cr["spec"].get("allocations") or {})
+            if n:
+                per_node[cr["metadata"]["name"]] = n
+    finally:
+        c.stop()
+    counts = sorted(per_node.values())
+    return {
+        "scenario": "manynode",
+        "policy": args.policy,
+        "nodes": n_nodes,
+        "pods": n_pods,
+        "pods_per_s": round(n_pods / wall, 2),
+        "latency": _lat_stats(lat),
+        "nodes_used": len(per_node),
+        "allocs_per_used_node": {"min": counts[0], "max": counts[-1]}
+        if counts else {},
+    }
+
+
 SCENARIOS = {
     "plumbing": scenario_plumbing,
+    "manynode": scenario_manynode,
     "flatlat": scenario_flatlat,
     "cpx8": scenario_cpx8,
     "mixed100": scenario_mixed100,

@@ -227,9 +227,12 @@ class PackedFitPolicy(AllocationPolicy):
        not per-slot.
     2. Among idle GPUs needing a mode change, prefer one whose memory mode
        already satisfies the profile (avoids the expensive NPS flip), then
-       lower physical index (keeps high-index GPUs free for large SPX jobs,
-       and co-locates small partitions on few GPUs so xGMI links of the
-       remaining GPUs stay uncontended for multi-GPU tenants).
+       a node that already hosts allocations (consolidate flips onto
+       already-busy nodes — without this term the cross-node argmax sends
+       every flip to a fresh node's GPU 0), then lower physical index
+       (keeps high-index GPUs free for large SPX jobs, and co-locates small
+       partitions on few GPUs so xGMI links of the remaining GPUs stay
+       uncontended for multi-GPU tenants).
     """
 
     name = "packed-fit"
@@ -242,19 +245,20 @@ class PackedFitPolicy(AllocationPolicy):
             if sp:
                 return sp
         best: Optional[Tuple[Tuple, Placement]] = None
+        node_used = sum(len(v.occupied) for v in views)
         for v in views:
             if v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
                 if p is None:
                     continue
                 # tier 2 = same mode; prefer most-occupied, then lowest index
-                score = (2, len(v.occupied), -v.index)
+                score = (2, len(v.occupied), 0, -v.index)
             else:
                 p = _place_on(v, profile, needs_change=True)
                 if p is None:
                     continue
                 mem_ok = v.memory_mode in VALID_MEMORY_MODES[profile.compute]
-                score = (1, 1 if mem_ok else 0, -v.index)
+                score = (1, 1 if mem_ok else 0, node_used, -v.index)
             if best is None or score > best[0]:
                 best = (score, p)
         return best

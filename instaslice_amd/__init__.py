@@ -46,3 +46,4 @@ POD_RESOURCE_PREFIX = "org.instaslice/"  # per-pod extended resource pinning
 # reference requeues silently, instaslice_controller.go:231; surfacing the
 # condition lets schedulers/users react instead of waiting blind)
 UNSCHEDULABLE_ANNOTATION = "org.instaslice/unschedulable"
+PRIORITY_ANNOTATION = "org.instaslice/priority"  # int; higher may preempt

@@ -65,6 +65,7 @@ class AllocationDetails:
     compute_mode: str = ""       # target compute mode, e.g. "CPX"
     memory_mode: str = ""        # target memory mode, e.g. "NPS4"
     group: str = ""              # gang label (org.instaslice/group annotation)
+    priority: int = 0            # org.instaslice/priority (preemption rank)
 
     def to_dict(self) -> dict:
         return {
@@ -81,6 +82,7 @@ class AllocationDetails:
             "computeMode": self.compute_mode,
             "memoryMode": self.memory_mode,
             "group": self.group,
+            "priority": self.priority,
         }
 
     @classmethod
@@ -99,6 +101,7 @@ class AllocationDetails:
             compute_mode=d.get("computeMode", ""),
             memory_mode=d.get("memoryMode", ""),
             group=d.get("group", ""),
+            priority=int(d.get("priority", 0)),
         )
 
 
@@ -216,6 +219,7 @@ def new_pod(
     gated: bool = True,
     group: Optional[str] = None,
     labels: Optional[Dict[str, str]] = None,
+    priority: int = 0,
 ) -> dict:
     """Synthetic gated pod following the reference's consumer contract
     (samples/test-pod.yaml:1-21): scheduling gate + finalizer + profile limit
@@ -236,9 +240,10 @@ def new_pod(
             "uid": uid or new_uid(),
             "finalizers": [FINALIZER_NAME],
             "deletionTimestamp": None,
-            "annotations": (
-                {"org.instaslice/group": group} if group else {}
-            ),
+            "annotations": {
+                **({"org.instaslice/group": group} if group else {}),
+                **({"org.instaslice/priority": str(priority)} if priority else {}),
+            },
             "labels": labels or {},
         },
         "spec": {

@@ -63,7 +63,11 @@ class GpuView:
     mode_locked: bool = False
 
 
-def build_gpu_views(node_name: str, spec: dict) -> List[GpuView]:
+NOMINATION_TTL_S = 30.0
+
+
+def build_gpu_views(node_name: str, spec: dict,
+                    nominee_uid: Optional[str] = None) -> List[GpuView]:
     """Assemble GpuViews from an Instaslice CR spec (api.types.new_instaslice
     shape). Sorted by physical index for deterministic first-fit.
 
@@ -84,6 +88,18 @@ def build_gpu_views(node_name: str, spec: dict) -> List[GpuView]:
             )
     for prep in (spec.get("prepared") or {}).values():
         occupied.setdefault(prep["parentGpuUUID"], set()).add(prep["ordinal"])
+    # preemption nominations: a slot freed by an eviction is reserved for
+    # the preemptor (k8s nominatedNodeName analog) — everyone else sees it
+    # occupied until the nominee lands or the reservation expires
+    import time as _time
+
+    now = _time.time()
+    for uid, nom in (spec.get("nominations") or {}).items():
+        if uid == nominee_uid:
+            continue
+        if now - float(nom.get("ts", 0)) > NOMINATION_TTL_S:
+            continue
+        occupied.setdefault(nom["gpuUUID"], set()).add(nom["ordinal"])
 
     views: List[GpuView] = []
     for uuid, gd in (spec.get("gpus") or {}).items():

@@ -94,6 +94,8 @@ class PodController:
         self._unschedulable_keys: set = set()
         # node -> (placements_dict, ProfileCatalog): see _profile_for
         self._catalog_cache: dict = {}
+        # pod uid -> (victim_name, t): preemption cooldown (_maybe_preempt)
+        self._preempted_for: dict = {}
         # workers > 1 is safe: the engine guarantees a key is never
         # reconciled concurrently (runtime/engine.py in-flight guard), and
         # placement races between different pods are caught by the
@@ -200,7 +202,8 @@ class PodController:
             profile = self._profile_for(cr, profile_name)
             if profile is None:
                 continue  # node does not offer this profile
-            views = build_gpu_views(node_name, cr.get("spec", {}))
+            views = build_gpu_views(node_name, cr.get("spec", {}),
+                                    nominee_uid=md["uid"])
             # gang affinity: GPUs already hosting this pod's group on this
             # node (same-GPU XCD co-location = intra-die bandwidth, no xGMI)
             prefer = frozenset(
@@ -214,6 +217,15 @@ class PodController:
             candidates.append((sp[0], cr, sp[1], profile))
             if self.policy.node_order_first_fit:
                 break
+        if not candidates:
+            # preemption (opt-in, org.instaslice/priority): evict the
+            # lowest-priority same-profile allocation with strictly lower
+            # priority; the drain frees an exact-fit slot and the existing
+            # unschedulable-retry machinery places this pod on the event.
+            # The reference has no preemption at all; k8s-style semantics
+            # (victim pod is DELETED, finalizer-grace applies).
+            if self._maybe_preempt(pod, profile_name, crs):
+                return Result(requeue_after=self.requeue_no_fit_s)
         # best score first; on a lost CAS race, fall through to the next node
         candidates.sort(key=lambda c: c[0], reverse=True)
         for _, cr, placement, profile in candidates:
@@ -232,6 +244,7 @@ class PodController:
                 compute_mode=placement.compute_mode,
                 memory_mode=placement.memory_mode,
                 group=group,
+                priority=self._pod_priority(pod),
             )
             updated = self._write_allocation(cr, node_name, alloc)
             if updated and md["uid"] in updated["spec"]["allocations"]:
@@ -285,7 +298,18 @@ class PodController:
         instaslice_controller.go:93)."""
         from instaslice_amd.store.memstore import Conflict
 
+        from instaslice_amd.controller.policy import NOMINATION_TTL_S
+
         a = alloc
+        # landing a placement consumes this pod's nomination (if any) and
+        # garbage-collects expired ones — valid at this CR revision, and the
+        # CAS guard makes the cleanup race-free
+        now = time.time()
+        nom_clears = [
+            {"op": "delete", "path": ["spec", "nominations", nuid]}
+            for nuid, nom in (cr["spec"].get("nominations") or {}).items()
+            if nuid == a.pod_uuid or now - float(nom.get("ts", 0)) > NOMINATION_TTL_S
+        ]
         try:
             return self.store.patch("Instaslice", node_name, INSTASLICE_NS, [
                 {"op": "test", "path": ["metadata", "resourceVersion"],
@@ -294,6 +318,7 @@ class PodController:
                  "absent": True},
                 {"op": "set", "path": ["spec", "allocations", a.pod_uuid],
                  "value": a.to_dict()},
+                *nom_clears,
             ])
         except NotFound:
             return None
@@ -305,9 +330,12 @@ class PodController:
             allocs = spec.setdefault("allocations", {})
             if a.pod_uuid in allocs:
                 return None  # raced with ourselves; done
+            spec.get("nominations", {}).pop(a.pod_uuid, None)
             # re-validate the placement against the fresh CR (another
-            # pod may have taken the ordinal between list() and now)
-            fresh_views = build_gpu_views(cr_obj["metadata"]["name"], spec)
+            # pod may have taken the ordinal between list() and now);
+            # nominee view so our own reserved slot doesn't read occupied
+            fresh_views = build_gpu_views(cr_obj["metadata"]["name"], spec,
+                                          nominee_uid=a.pod_uuid)
             for v in fresh_views:
                 if v.uuid == a.gpu_uuid:
                     if a.ordinal in v.occupied:
@@ -323,6 +351,77 @@ class PodController:
         return self.store.update_with_retry(
             "Instaslice", node_name, INSTASLICE_NS, add_alloc
         )
+
+    @staticmethod
+    def _pod_priority(pod: dict) -> int:
+        from instaslice_amd import PRIORITY_ANNOTATION
+
+        try:
+            return int((pod["metadata"].get("annotations") or {})
+                       .get(PRIORITY_ANNOTATION, 0))
+        except (TypeError, ValueError):
+            return 0
+
+    def _maybe_preempt(self, pod: dict, profile_name: str,
+                       crs: List[dict]) -> bool:
+        """Evict ONE lowest-priority, strictly-lower-priority allocation of
+        the SAME profile (exact slot fit; cross-profile/mode-flip preemption
+        is intentionally out of scope — it would need multi-victim planning).
+        Returns True if an eviction was issued. A per-pod cooldown prevents
+        cascading evictions while the first victim is still draining."""
+        prio = self._pod_priority(pod)
+        if prio <= 0:
+            return False
+        uid = pod["metadata"]["uid"]
+        last = self._preempted_for.get(uid)
+        if last is not None and time.monotonic() - last[1] < 10.0:
+            # victim still draining (its DELETED event will retrigger us)
+            return False
+        victim = None  # (priority, cr_name, alloc_dict)
+        for cr in crs:
+            if self._node_stale(cr):
+                continue
+            for a in (cr["spec"].get("allocations") or {}).values():
+                if a.get("profile") != profile_name:
+                    continue
+                if a.get("allocationStatus") == AllocationStatus.DELETED:
+                    return False  # a slot is already draining; just wait
+                vprio = int(a.get("priority", 0))
+                if vprio >= prio:
+                    continue
+                if victim is None or vprio < victim[0]:
+                    victim = (vprio, cr["metadata"]["name"], a)
+        if victim is None:
+            return False
+        vprio, v_node, v_alloc = victim
+        try:
+            self.store.delete("Pod", v_alloc["podName"], v_alloc["namespace"])
+        except NotFound:
+            return False
+        # reserve the freed slot for THIS pod (k8s nominatedNodeName analog):
+        # without it, any waiting pod can steal the slot when it drains and
+        # the preemptor evicts again — a cascade (caught by the e2e test)
+        try:
+            self.store.patch("Instaslice", v_node, INSTASLICE_NS, [
+                {"op": "set", "path": ["spec", "nominations", uid],
+                 "value": {"gpuUUID": v_alloc["gpuUUID"],
+                           "ordinal": v_alloc["ordinal"],
+                           "ts": time.time()}},
+            ], quiet=True)
+        except NotFound:
+            pass
+        self._preempted_for[uid] = (v_alloc["podName"], time.monotonic())
+        self.log.warning(
+            "preempting pod %s/%s (priority %d) for %s (priority %d); slot "
+            "%s#%d nominated",
+            v_alloc["namespace"], v_alloc["podName"], vprio,
+            pod["metadata"]["name"], prio,
+            v_alloc["gpuUUID"][:8], v_alloc["ordinal"],
+        )
+        from instaslice_amd.metrics import get_metrics
+
+        get_metrics().allocation("preempted")
+        return True
 
     def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str,
                                expect: Optional[str] = None) -> None:

@@ -407,3 +407,39 @@ def test_reconfigure_bumps_device_plugin_label():
             "org.instaslice/last-reconfigure")
     finally:
         c.stop()
+
+
+def test_priority_preemption_evicts_lowest():
+    """A full node + a higher-priority pod: the LOWEST-priority same-profile
+    pod is evicted (k8s-preemption analog; opt-in via org.instaslice/priority
+    annotation — the reference has no preemption). Equal/higher-priority
+    pods are never victims."""
+    from instaslice_amd.api.types import new_pod
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        # fill all 8 CPX slots at priority 5, except low0 at priority 2
+        for i in range(8):
+            c.store.create(new_pod(f"low{i}", profile="cpx-1x36",
+                                   priority=2 if i == 0 else 5))
+            c.wait_pod_scheduled(f"low{i}")
+        # equal priority NEVER preempts (strictly-lower victims only...
+        # priority 2 < 5 though — so use priority 2 contender first)
+        c.store.create(new_pod("equal", profile="cpx-1x36", priority=2))
+        assert c.wait_pod_outcome("equal") == "unschedulable"
+        pods = {p["metadata"]["name"] for p in c.store.list("Pod")}
+        assert {f"low{i}" for i in range(8)} <= pods
+        # priority 10 preempts the LOWEST victim (low0 at priority 2)
+        c.store.create(new_pod("high", profile="cpx-1x36", priority=10))
+        c.wait_pod_scheduled("high", timeout=15.0)
+        pods = {p["metadata"]["name"] for p in c.store.list("Pod")}
+        assert "low0" not in pods, "lowest-priority victim not evicted"
+        assert {f"low{i}" for i in range(1, 8)} <= pods
+        # ... and the lower-priority contender is still waiting, gated
+        assert "equal" in pods
+        eq = c.store.get("Pod", "equal", "default")
+        assert eq["spec"].get("schedulingGates")
+    finally:
+        c.stop()

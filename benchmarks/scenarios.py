@@ -316,9 +316,33 @@ def scenario_manynode(args) -> dict:
     }
 
 
+def scenario_preempt(args) -> dict:
+    """Preemption latency: node full of low-priority pods; measure
+    time-to-scheduled for high-priority arrivals (eviction + drain +
+    nominated re-place, org.instaslice/priority)."""
+    from instaslice_amd.api.types import new_pod
+
+    c = _mk_cluster(num_gpus=1)
+    lat = []
+    try:
+        for i in range(8):
+            c.store.create(new_pod(f"lo-{i}", profile="cpx-1x36", priority=1))
+            c.wait_pod_scheduled(f"lo-{i}")
+        for i in range(args.pods or 8):
+            t0 = time.perf_counter()
+            c.store.create(new_pod(f"hi-{i}", profile="cpx-1x36",
+                                   priority=10))
+            c.wait_pod_scheduled(f"hi-{i}", timeout=30.0)
+            lat.append((time.perf_counter() - t0) * 1000)
+    finally:
+        c.stop()
+    return {"scenario": "preempt", "preemption_latency": _lat_stats(lat)}
+
+
 SCENARIOS = {
     "plumbing": scenario_plumbing,
     "manynode": scenario_manynode,
+    "preempt": scenario_preempt,
     "flatlat": scenario_flatlat,
     "cpx8": scenario_cpx8,
     "mixed100": scenario_mixed100,

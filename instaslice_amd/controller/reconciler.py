@@ -225,6 +225,11 @@ class PodController:
             # The reference has no preemption at all; k8s-style semantics
             # (victim pod is DELETED, finalizer-grace applies).
             if self._maybe_preempt(pod, profile_name, crs):
+                # register for event-driven retry: the victim's drain emits
+                # CR events that re-reconcile unschedulable pods, landing
+                # the preemptor on its nominated slot within milliseconds
+                # (the timer below is only the dead-agent fallback)
+                self._mark_unschedulable(md, profile_name, unschedulable=True)
                 return Result(requeue_after=self.requeue_no_fit_s)
         # best score first; on a lost CAS race, fall through to the next node
         candidates.sort(key=lambda c: c[0], reverse=True)

@@ -347,3 +347,87 @@ def test_native_persistence_checkpoint_resume(tmp_path):
     finally:
         c.close()
         server.stop()
+
+
+@needs_stored
+def test_chaos_store_restart_mid_churn(tmp_path):
+    """Kill (SIGTERM -> snapshot flush) and restart the persistent store
+    daemon WHILE pods are churning: reconnect-enabled controller and agent
+    resync their watches, state resumes from the snapshot, and the churn
+    completes with consistent allocations. This is the etcd-restart drill
+    the reference delegates to Kubernetes."""
+    from instaslice_amd.agent.daemonset import NodeAgent
+    from instaslice_amd.controller.reconciler import PodController
+    from instaslice_amd.smi.fake import FakeAmdSmi
+
+    snap = str(tmp_path / "chaos.msgpack")
+    server = NativeStoreServer(persist_path=snap).start()
+    port = server.port
+    cstore = NetStoreClient("127.0.0.1", port, reconnect=True)
+    astore = NetStoreClient("127.0.0.1", port, reconnect=True)
+    controller = PodController(cstore, teardown_grace_s=0.0, workers=2)
+    controller.requeue_no_fit_s = 0.05
+    agent = NodeAgent(astore, FakeAmdSmi(num_gpus=2, node_name="node-0"),
+                      "node-0", heartbeat_every_s=0.5)
+    agent.start()
+    controller.start()
+    bench = NetStoreClient("127.0.0.1", port, reconnect=True)
+
+    def wait_sched(name, timeout=30.0):
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            try:
+                p = bench.get("Pod", name, "default")
+                if not p["spec"].get("schedulingGates"):
+                    return
+            except (NotFound, ConnectionError, TimeoutError, RuntimeError):
+                pass
+            time.sleep(0.05)
+        raise TimeoutError(name)
+
+    try:
+        for i in range(4):
+            bench.create(new_pod(f"pre{i}", profile="cpx-1x36"))
+            wait_sched(f"pre{i}")
+        # hard restart: flush + new daemon on the same port/path
+        server.stop()
+        time.sleep(0.5)
+        server = NativeStoreServer(port=port, persist_path=snap).start()
+        # pre-restart state resumed?
+        deadline = time.monotonic() + 20
+        while time.monotonic() < deadline:
+            try:
+                cr = bench.get("Instaslice", "node-0", "instaslice-system")
+                break
+            except (NotFound, ConnectionError, TimeoutError, RuntimeError):
+                time.sleep(0.2)
+        else:
+            pytest.fail("CR not resumed after restart")
+        assert len(cr["spec"]["allocations"]) == 4
+        # churn continues against the resumed store
+        for i in range(6):
+            created = False
+            deadline = time.monotonic() + 20
+            while time.monotonic() < deadline:
+                try:
+                    bench.create(new_pod(f"post{i}", profile="cpx-1x36"))
+                    created = True
+                    break
+                except (ConnectionError, TimeoutError, RuntimeError):
+                    time.sleep(0.2)
+                except AlreadyExists:
+                    created = True
+                    break
+            assert created
+            wait_sched(f"post{i}")
+        cr = bench.get("Instaslice", "node-0", "instaslice-system")
+        allocs = cr["spec"]["allocations"]
+        assert len(allocs) == 10
+        slots = {(a["gpuUUID"], a["ordinal"]) for a in allocs.values()}
+        assert len(slots) == 10, "double-booked slot after restart"
+    finally:
+        controller.stop()
+        agent.stop()
+        for c in (cstore, astore, bench):
+            c.close()
+        server.stop()

@@ -239,6 +239,53 @@ def cmd_status(args) -> int:
     return 0
 
 
+def cmd_top(args) -> int:
+    """Live cluster view driven by store watches (kubectl-get -w analog):
+    one line per state change, plus a periodic occupancy summary."""
+    store = _connect(args.store)
+    w = store.watch(replay=True, filters=[{"kind": "Pod"},
+                                          {"kind": "Instaslice"}])
+    last_summary = 0.0
+    try:
+        while True:
+            ev = w.next(timeout=1.0)
+            now = time.monotonic()
+            if ev is not None:
+                et, obj = ev
+                md = obj["metadata"]
+                if obj["kind"] == "Pod":
+                    gates = "gated" if obj["spec"].get("schedulingGates") else "ungated"
+                    print(f"{et:9s} pod {md.get('namespace','')}/{md['name']} {gates}",
+                          flush=True)
+                else:
+                    spec = obj.get("spec", {})
+                    allocs = spec.get("allocations") or {}
+                    print(f"{et:9s} instaslice {md['name']} "
+                          f"allocs={len(allocs)} "
+                          f"prepared={len(spec.get('prepared') or {})}",
+                          flush=True)
+            if now - last_summary > args.interval:
+                last_summary = now
+                for cr in store.list("Instaslice"):
+                    spec = cr.get("spec", {})
+                    used = sum(len(g.get("usedOrdinals", []))
+                               for g in (spec.get("gpus") or {}).values())
+                    total = sum(
+                        {"SPX": 1, "DPX": 2, "TPX": 3, "QPX": 4, "CPX": 8}
+                        .get(g.get("computeMode", "SPX"), 1)
+                        for g in (spec.get("gpus") or {}).values())
+                    print(f"--- {cr['metadata']['name']}: {used}/{total} "
+                          f"partitions used, "
+                          f"{len(spec.get('allocations') or {})} allocations",
+                          flush=True)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        w.stop()
+        store.close()
+    return 0
+
+
 def cmd_payload(args) -> int:
     import subprocess
 
@@ -317,6 +364,11 @@ def main(argv=None) -> int:
     p = sub.add_parser("status", help="cluster state summary")
     p.add_argument("--store", default="127.0.0.1:7080")
     p.set_defaults(fn=cmd_status)
+
+    p = sub.add_parser("top", help="live cluster view (watch-driven)")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--interval", type=float, default=10.0)
+    p.set_defaults(fn=cmd_top)
 
     p = sub.add_parser("payload", help="run the HIP payload binary")
     p.add_argument("payload_args", nargs="*", default=["info"])

@@ -66,6 +66,12 @@ class GpuView:
 NOMINATION_TTL_S = 30.0
 
 
+def _now() -> float:
+    import time
+
+    return time.time()
+
+
 def build_gpu_views(node_name: str, spec: dict,
                     nominee_uid: Optional[str] = None) -> List[GpuView]:
     """Assemble GpuViews from an Instaslice CR spec (api.types.new_instaslice
@@ -91,9 +97,7 @@ def build_gpu_views(node_name: str, spec: dict,
     # preemption nominations: a slot freed by an eviction is reserved for
     # the preemptor (k8s nominatedNodeName analog) — everyone else sees it
     # occupied until the nominee lands or the reservation expires
-    import time as _time
-
-    now = _time.time()
+    now = _now()
     for uid, nom in (spec.get("nominations") or {}).items():
         if uid == nominee_uid:
             continue

@@ -341,6 +341,14 @@ class NodeAgent:
             # that change re-reconciles us with fresh state
             self.log.debug("prepare commit superseded: %s", commit["error"]["msg"])
             return False
+        from instaslice_amd.api.events import emit
+
+        emit(self.store,
+             {"kind": "Pod", "namespace": alloc["namespace"],
+              "name": alloc["podName"]},
+             "PartitionReady",
+             f"partition {part_uuid[:8]} (ordinal {alloc['ordinal']} on "
+             f"{gpu_uuid[:8]}, device {prep['deviceIndex']}) realized")
         return True
 
     def _needs_mode_change(self, alloc: dict) -> bool:
@@ -424,6 +432,14 @@ class NodeAgent:
                         alloc["podName"], gpu[:8], e,
                     )
                     self._mode_set_failures.pop(gpu, None)
+                    from instaslice_amd.api.events import emit
+
+                    emit(self.store,
+                         {"kind": "Pod", "namespace": alloc["namespace"],
+                          "name": alloc["podName"]},
+                         "PartitionFailed",
+                         f"device error realizing partition on "
+                         f"{gpu[:8]}: {e}", type_="Warning")
                     self._fail_allocation(pod_uuid, gpu,
                                           lock_mode=deterministic and needs_flip)
                     continue
@@ -484,6 +500,14 @@ class NodeAgent:
             from instaslice_amd.metrics import get_metrics
 
             get_metrics().allocation("deleted")
+            from instaslice_amd.api.events import emit
+
+            emit(self.store,
+                 {"kind": "Pod", "namespace": alloc["namespace"],
+                  "name": alloc["podName"]},
+                 "PartitionReleased",
+                 f"partition ordinal {alloc['ordinal']} on "
+                 f"{alloc['gpuUUID'][:8]} released")
 
     def _maybe_reset_gpu(self, gpu_uuid: str) -> None:
         """Reference-parity teardown (ci/gi Destroy analog,

@@ -1,0 +1,155 @@
+"""Kubernetes-style Event emission (kubectl-describe analog).
+
+The reference emits no Events (controller-runtime recorders unused); real
+operators do, and they are the first thing a cluster admin reads when a pod
+sits gated. Events here are ordinary store objects (kind "Event") keyed
+`<involved-name>.<reason>.<n>` with k8s-ish dedup: re-emitting the same
+(involved, reason) bumps `count` and `lastTimestamp` via a single PATCH
+instead of creating a new object.
+
+Usage: emit(store, involved_obj_or_ref, reason, message, type_="Normal").
+Failures never propagate — events are best-effort observability."""
+
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+from instaslice_amd.store.memstore import AlreadyExists, Conflict, NotFound
+from instaslice_amd.utils import get_logger
+
+log = get_logger("events")
+
+
+def _ref(obj: dict) -> dict:
+    md = obj.get("metadata", {})
+    return {
+        "kind": obj.get("kind", ""),
+        "namespace": md.get("namespace", ""),
+        "name": md.get("name", ""),
+        "uid": md.get("uid", ""),
+    }
+
+
+class _EventSink:
+    """Asynchronous recorder (k8s EventRecorder analog): emit() is a queue
+    put; a single daemon thread does the store writes, so events never sit
+    on the allocation critical path (measured ~20% bench throughput when
+    synchronous). Bounded queue; overflow drops the newest and counts it."""
+
+    # retention: events are keyed per (involved, reason), so a churning
+    # cluster mints one forever-object per pod. Each sink caps what IT
+    # created with an LRU (k8s uses a 1 h etcd TTL; a store-side TTL would
+    # need a sweeper — per-writer LRU is cheaper and local)
+    MAX_LIVE_EVENTS = 2000
+
+    def __init__(self) -> None:
+        import collections
+        import queue
+        import threading
+
+        self._q: "queue.Queue" = queue.Queue(maxsize=4096)
+        self._lru: "collections.OrderedDict" = collections.OrderedDict()
+        self.dropped = 0
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="event-sink")
+        self._thread.start()
+
+    def put(self, item) -> None:
+        try:
+            self._q.put_nowait(item)
+        except Exception:
+            self.dropped += 1
+
+    def _run(self) -> None:
+        while True:
+            store, ref, ns, name, reason, message, type_, component = self._q.get()
+            try:
+                _write_event(store, ref, ns, name, reason, message,
+                             type_, component)
+                self._lru[(ns, name)] = store
+                self._lru.move_to_end((ns, name))
+                while len(self._lru) > self.MAX_LIVE_EVENTS:
+                    (old_ns, old_name), old_store = self._lru.popitem(last=False)
+                    try:
+                        old_store.delete("Event", old_name, old_ns)
+                    except Exception:  # noqa: BLE001 - already gone is fine
+                        pass
+            except Exception as e:  # noqa: BLE001 - best-effort
+                log.debug("event write failed: %s", e)
+
+    def flush(self, timeout: float = 2.0) -> None:
+        """Best-effort drain (tests)."""
+        deadline = time.time() + timeout
+        while not self._q.empty() and time.time() < deadline:
+            time.sleep(0.01)
+
+
+_sink: Optional[_EventSink] = None
+
+
+def _get_sink() -> _EventSink:
+    global _sink
+    if _sink is None:
+        _sink = _EventSink()
+    return _sink
+
+
+def flush(timeout: float = 2.0) -> None:
+    if _sink is not None:
+        _sink.flush(timeout)
+
+
+def emit(store, involved: dict, reason: str, message: str,
+         type_: str = "Normal", component: str = "instaslice",
+         namespace: Optional[str] = None) -> None:
+    """Record (or dedup-bump) an Event for `involved` (an object or a ref
+    dict with kind/namespace/name). Asynchronous: returns immediately."""
+    ref = _ref(involved) if "metadata" in involved else dict(involved)
+    ns = namespace if namespace is not None else (ref.get("namespace") or "default")
+    name = f"{ref.get('name', '?')}.{reason}"
+    _get_sink().put((store, ref, ns, name, reason, message, type_, component))
+
+
+def _write_event(store, ref: dict, ns: str, name: str, reason: str,
+                 message: str, type_: str, component: str) -> None:
+    now = time.time()
+    # dedup bump needs read-modify; guarded patch, capped retries — events
+    # are best-effort
+    for _ in range(2):
+        try:
+            cur = store.get("Event", name, ns)
+        except NotFound:
+            ev = {
+                "apiVersion": "v1",
+                "kind": "Event",
+                "metadata": {"name": name, "namespace": ns},
+                "involvedObject": ref,
+                "reason": reason,
+                "message": message,
+                "type": type_,
+                "source": {"component": component},
+                "count": 1,
+                "firstTimestamp": now,
+                "lastTimestamp": now,
+            }
+            try:
+                store.create(ev)
+            except AlreadyExists:
+                continue
+            except Exception as e:  # noqa: BLE001
+                log.debug("event create failed: %s", e)
+            return
+        try:
+            store.patch("Event", name, ns, [
+                {"op": "test", "path": ["count"], "value": cur.get("count", 1)},
+                {"op": "set", "path": ["count"], "value": cur.get("count", 1) + 1},
+                {"op": "set", "path": ["lastTimestamp"], "value": now},
+                {"op": "set", "path": ["message"], "value": message},
+            ], quiet=True)
+            return
+        except Conflict:
+            continue
+        except Exception as e:  # noqa: BLE001
+            log.debug("event bump failed: %s", e)
+            return

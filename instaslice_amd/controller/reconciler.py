@@ -258,12 +258,24 @@ class PodController:
                     md["name"], profile_name, node_name,
                     placement.gpu_uuid[:8], placement.ordinal,
                 )
+                from instaslice_amd.api.events import emit
+
+                emit(self.store, pod, "Placed",
+                     f"allocated {profile_name} on {node_name}/"
+                     f"{placement.gpu_uuid[:8]}#{placement.ordinal}"
+                     + (" (mode change planned)"
+                        if placement.needs_mode_change else ""))
                 self._mark_unschedulable(md, profile_name, unschedulable=False)
                 # normally the agent's "created" event advances this pod; the
                 # timed recheck only notices a dead agent (no events then)
                 return Result(
                     requeue_after=self.node_stale_after_s or None
                 )
+        from instaslice_amd.api.events import emit
+
+        emit(self.store, pod, "Unschedulable",
+             f"no node can host profile {profile_name} right now",
+             type_="Warning")
         self._mark_unschedulable(md, profile_name, unschedulable=True)
         return Result(requeue_after=self.requeue_no_fit_s)
 
@@ -415,6 +427,17 @@ class PodController:
             ], quiet=True)
         except NotFound:
             pass
+        from instaslice_amd.api.events import emit
+
+        emit(self.store, pod, "Preempting",
+             f"evicting {v_alloc['namespace']}/{v_alloc['podName']} "
+             f"(priority {vprio}) and nominating its slot", type_="Warning")
+        emit(self.store,
+             {"kind": "Pod", "namespace": v_alloc["namespace"],
+              "name": v_alloc["podName"]},
+             "Preempted",
+             f"evicted for higher-priority pod {pod['metadata']['name']} "
+             f"(priority {prio})", type_="Warning")
         self._preempted_for[uid] = (v_alloc["podName"], time.monotonic())
         self.log.warning(
             "preempting pod %s/%s (priority %d) for %s (priority %d); slot "

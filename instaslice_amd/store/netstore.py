@@ -319,11 +319,14 @@ class NetStoreClient:
                 self._sock = sock
                 self._rfile = sock.makefile("rb")
             self._fail_pending("connection lost")  # sent into the dead socket
+            # NOTE: _watch_specs must survive a FAILED resubscribe attempt
+            # (e.g. connecting to the dying server's last accept) — it is
+            # only replaced after every watch re-registered successfully
             with self._watch_reg_lock:
                 live = [(w, spec) for (w, spec) in self._watch_specs.values()
                         if not w._stopped]
-                self._watches = {}
-                self._watch_specs = {}
+                self._watches = {}  # old wids are invalid on the new conn
+            new_specs = {}
             ok = True
             for w, spec in live:
                 with self._idlock:
@@ -343,7 +346,7 @@ class NetStoreClient:
                             new_wid = msg["result"]["watch_id"]
                             with self._watch_reg_lock:
                                 self._watches[new_wid] = w
-                                self._watch_specs[new_wid] = (w, spec)
+                            new_specs[new_wid] = (w, spec)
                             break
                         self._dispatch(msg)
                 except (OSError, ValueError, KeyError):
@@ -351,10 +354,15 @@ class NetStoreClient:
                 if not ok:
                     break
             if ok:
+                with self._watch_reg_lock:
+                    self._watch_specs = new_specs
                 self.log.warning(
                     "reconnected to store %s:%d (%d watches resynced)",
                     self.host, self.port, len(live))
                 return True
+            # failed attempt: drop partial registrations, keep specs, retry
+            with self._watch_reg_lock:
+                self._watches = {}
         return False
 
     def _call(self, verb: str, **kw):

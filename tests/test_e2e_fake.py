@@ -443,3 +443,45 @@ def test_priority_preemption_evicts_lowest():
         assert eq["spec"].get("schedulingGates")
     finally:
         c.stop()
+
+
+def test_events_emitted_through_lifecycle():
+    """kubectl-describe analog: Placed/PartitionReady on the happy path,
+    Unschedulable (Warning, deduped with count) when full, Preempting/
+    Preempted around an eviction. The reference emits no Events at all."""
+    from instaslice_amd.api.types import new_pod
+
+    import time as _t
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+
+    def wait_event(name, timeout=5.0):
+        # events are recorded ASYNCHRONOUSLY (api/events._EventSink)
+        deadline = _t.monotonic() + timeout
+        while _t.monotonic() < deadline:
+            try:
+                return c.store.get("Event", name, "default")
+            except Exception:
+                _t.sleep(0.02)
+        raise TimeoutError(f"event {name} never recorded")
+
+    try:
+        c.submit_pod("ev0", "spx-8x288")
+        c.wait_pod_scheduled("ev0")
+        assert wait_event("ev0.Placed")["involvedObject"]["name"] == "ev0"
+        wait_event("ev0.PartitionReady")
+        # capacity full -> Unschedulable warning, deduped on retries
+        c.submit_pod("ev1", "spx-8x288")
+        assert c.wait_pod_outcome("ev1") == "unschedulable"
+        ev = wait_event("ev1.Unschedulable")
+        assert ev["type"] == "Warning" and ev["count"] >= 1
+        # preemption pair
+        c.store.create(new_pod("hi", profile="spx-8x288", priority=9))
+        c.wait_pod_scheduled("hi", timeout=15.0)
+        wait_event("hi.Preempting")
+        wait_event("ev0.Preempted")
+        wait_event("ev0.PartitionReleased")
+    finally:
+        c.stop()

@@ -239,6 +239,40 @@ def cmd_status(args) -> int:
     return 0
 
 
+def cmd_describe(args) -> int:
+    """Pod detail + its Events (kubectl-describe analog)."""
+    store = _connect(args.store)
+    out = {}
+    try:
+        pod = store.get("Pod", args.name, args.namespace)
+        out["pod"] = {
+            "name": pod["metadata"]["name"],
+            "namespace": pod["metadata"].get("namespace", ""),
+            "gated": bool(pod["spec"].get("schedulingGates")),
+            "annotations": pod["metadata"].get("annotations") or {},
+            "limits": (pod["spec"]["containers"][0]["resources"]
+                       .get("limits", {})),
+        }
+    except Exception as e:  # noqa: BLE001
+        out["pod"] = f"not found: {e}"
+    for cr in store.list("Instaslice"):
+        for a in (cr["spec"].get("allocations") or {}).values():
+            if a["podName"] == args.name and a["namespace"] == args.namespace:
+                out["allocation"] = a
+    out["events"] = sorted(
+        (
+            {"reason": e["reason"], "type": e["type"], "count": e["count"],
+             "message": e["message"], "lastTimestamp": e["lastTimestamp"]}
+            for e in store.list("Event", args.namespace)
+            if e.get("involvedObject", {}).get("name") == args.name
+        ),
+        key=lambda e: e["lastTimestamp"],
+    )
+    print(json.dumps(out, indent=2))
+    store.close()
+    return 0
+
+
 def cmd_top(args) -> int:
     """Live cluster view driven by store watches (kubectl-get -w analog):
     one line per state change, plus a periodic occupancy summary."""
@@ -364,6 +398,12 @@ def main(argv=None) -> int:
     p = sub.add_parser("status", help="cluster state summary")
     p.add_argument("--store", default="127.0.0.1:7080")
     p.set_defaults(fn=cmd_status)
+
+    p = sub.add_parser("describe", help="pod detail + events")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--name", required=True)
+    p.add_argument("--namespace", default="default")
+    p.set_defaults(fn=cmd_describe)
 
     p = sub.add_parser("top", help="live cluster view (watch-driven)")
     p.add_argument("--store", default="127.0.0.1:7080")

@@ -55,7 +55,15 @@ class _EventSink:
                                         name="event-sink")
         self._thread.start()
 
+    # under sustained churn Normal events are sampled away once the queue
+    # backs up (k8s recorders rate-limit the same way); Warnings always keep
+    NORMAL_SHED_DEPTH = 256
+
     def put(self, item) -> None:
+        type_ = item[6]
+        if type_ == "Normal" and self._q.qsize() > self.NORMAL_SHED_DEPTH:
+            self.dropped += 1
+            return
         try:
             self._q.put_nowait(item)
         except Exception:

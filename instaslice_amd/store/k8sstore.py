@@ -116,6 +116,9 @@ class K8sStore:
                 return self._to_dict(self._core.create_node(obj))
             if kind == "Lease":
                 return self._to_dict(self._coord.create_namespaced_lease(ns, obj))
+            if kind == "Event":
+                return self._to_dict(
+                    self._core.create_namespaced_event(ns, obj))
             raise ValueError(f"unsupported kind {kind}")
         except ApiException as e:
             self._raise(e, kind, name)
@@ -135,6 +138,9 @@ class K8sStore:
                 return self._to_dict(self._core.read_node(name))
             if kind == "Lease":
                 return self._to_dict(self._coord.read_namespaced_lease(name, ns))
+            if kind == "Event":
+                return self._to_dict(
+                    self._core.read_namespaced_event(name, ns))
             raise ValueError(f"unsupported kind {kind}")
         except ApiException as e:
             self._raise(e, kind, name)
@@ -181,6 +187,9 @@ class K8sStore:
             if kind == "Lease":
                 return self._to_dict(
                     self._coord.replace_namespaced_lease(name, ns, obj))
+            if kind == "Event":
+                return self._to_dict(
+                    self._core.replace_namespaced_event(name, ns, obj))
             raise ValueError(f"unsupported kind {kind}")
         except ApiException as e:
             self._raise(e, kind, name)
@@ -198,6 +207,8 @@ class K8sStore:
                 self._core.delete_namespaced_config_map(name, ns)
             elif kind == "Node":
                 self._core.delete_node(name)
+            elif kind == "Event":
+                self._core.delete_namespaced_event(name, ns)
             else:
                 raise ValueError(f"unsupported kind {kind}")
         except ApiException as e:

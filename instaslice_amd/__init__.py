@@ -47,3 +47,4 @@ POD_RESOURCE_PREFIX = "org.instaslice/"  # per-pod extended resource pinning
 # condition lets schedulers/users react instead of waiting blind)
 UNSCHEDULABLE_ANNOTATION = "org.instaslice/unschedulable"
 PRIORITY_ANNOTATION = "org.instaslice/priority"  # int; higher may preempt
+GROUP_SIZE_ANNOTATION = "org.instaslice/group-size"  # gang ungate barrier

@@ -369,6 +369,34 @@ class PodController:
             "Instaslice", node_name, INSTASLICE_NS, add_alloc
         )
 
+    def _gang_ready(self, pod: dict, alloc: dict) -> bool:
+        from instaslice_amd import GROUP_SIZE_ANNOTATION
+
+        ann = pod["metadata"].get("annotations") or {}
+        group = alloc.get("group") or ""
+        try:
+            want = int(ann.get(GROUP_SIZE_ANNOTATION, 0))
+        except (TypeError, ValueError):
+            want = 0
+        if not group or want <= 1:
+            return True
+        have = sum(
+            1
+            for cr in self._crs()
+            for a in (cr["spec"].get("allocations") or {}).values()
+            if a.get("group") == group
+            and a.get("allocationStatus") in (AllocationStatus.CREATED,
+                                              AllocationStatus.UNGATED)
+        )
+        if have < want:
+            from instaslice_amd.api.events import emit
+
+            emit(self.store, pod, "GangWaiting",
+                 f"partition ready; waiting for gang '{group}' "
+                 f"({have}/{want} realized)")
+            return False
+        return True
+
     @staticmethod
     def _pod_priority(pod: dict) -> int:
         from instaslice_amd import PRIORITY_ANNOTATION
@@ -582,6 +610,15 @@ class PodController:
             )
             return Result(requeue_after=0.01)
         if status == AllocationStatus.CREATED:
+            # gang ungate barrier (org.instaslice/group + group-size):
+            # members keep their realized partitions but only ungate once
+            # the WHOLE gang is realized — the RCCL-workload contract (all
+            # ranks must start together; SURVEY.md §5 xGMI co-placement).
+            # Event-driven: every member with a `created` allocation is
+            # re-reconciled on each CR change, so the barrier lifts on the
+            # event that realizes the last member.
+            if not self._gang_ready(pod, alloc):
+                return Result()
             # agent realized the partition: let the pod schedule. One batched
             # round-trip: ungate the pod AND flip the allocation to ungated.
             from instaslice_amd import GATE_NAME

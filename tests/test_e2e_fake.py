@@ -485,3 +485,44 @@ def test_events_emitted_through_lifecycle():
         wait_event("ev0.PartitionReleased")
     finally:
         c.stop()
+
+
+def test_gang_ungate_barrier():
+    """group + group-size: members hold realized partitions but ungate
+    TOGETHER once the whole gang is ready (RCCL all-ranks-start contract).
+    A member that cannot be placed keeps the others gated."""
+    from instaslice_amd.api.types import new_pod
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+
+    def gang_pod(name):
+        p = new_pod(name, profile="cpx-1x36", group="ring")
+        p["metadata"]["annotations"]["org.instaslice/group-size"] = "3"
+        return p
+
+    try:
+        c.store.create(gang_pod("g0"))
+        c.store.create(gang_pod("g1"))
+        import time as _t
+
+        deadline = _t.monotonic() + 10
+        while _t.monotonic() < deadline:
+            cr = c.store.get("Instaslice", "node-0", "instaslice-system")
+            sts = sorted(a["allocationStatus"]
+                         for a in cr["spec"]["allocations"].values())
+            if sts == ["created", "created"]:
+                break
+            _t.sleep(0.02)
+        assert sts == ["created", "created"], sts
+        # partitions are realized but pods stay gated (barrier holds)
+        _t.sleep(0.3)
+        for n in ("g0", "g1"):
+            assert c.store.get("Pod", n, "default")["spec"]["schedulingGates"]
+        # third member completes the gang -> all three ungate
+        c.store.create(gang_pod("g2"))
+        for n in ("g0", "g1", "g2"):
+            c.wait_pod_scheduled(n, timeout=10.0)
+    finally:
+        c.stop()

@@ -287,6 +287,14 @@ def main() -> int:
     log.info("rank %d: backend=%s partitioning=%s profiles=%s payload=%s",
              rank, backend_name, partitioning, profiles, payload_on)
 
+    # GC latency hygiene for the agent+bench process (control-plane
+    # processes tune themselves in runtime/controlplane.py)
+    import gc
+
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(50000, 50, 50)
+
     def sync():
         # no in-process GPU work to drain (by design — see module docstring);
         # guard keeps the contract call without creating a HIP context

@@ -17,6 +17,18 @@ import sys
 from typing import Optional
 
 
+def _tune_gc() -> None:
+    """Latency hygiene: gen-0 collections at the default threshold (700)
+    pause the reconcile path every few thousand allocations; freeze the
+    startup object graph and raise the thresholds so collections are rare
+    and cheap. (Daemons keep GC ENABLED — only the cadence changes.)"""
+    import gc
+
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(50000, 50, 50)
+
+
 def _run_controller_shard(
     port: int,
     policy: str,
@@ -29,6 +41,7 @@ def _run_controller_shard(
 ) -> None:
     """multiprocessing target: one controller shard over TCP."""
     sys.setswitchinterval(0.001)
+    _tune_gc()
     from instaslice_amd.controller.reconciler import PodController
     from instaslice_amd.store.netstore import NetStoreClient
 
@@ -79,6 +92,7 @@ def run_control_plane(
     Both support persist_path (checkpoint/resume)."""
     # short GIL switch interval: the reconcile path is wakeup-latency bound
     sys.setswitchinterval(0.001)
+    _tune_gc()
     from instaslice_amd.controller.reconciler import PodController
     from instaslice_amd.store.memstore import MemStore
     from instaslice_amd.store.native import NativeStoreServer, stored_available

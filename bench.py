@@ -287,6 +287,15 @@ def main() -> int:
     log.info("rank %d: backend=%s partitioning=%s profiles=%s payload=%s",
              rank, backend_name, partitioning, profiles, payload_on)
 
+    prof_dir = os.environ.get("INSTASLICE_PROFILE_DIR")
+    prof_stop = None
+    if prof_dir:
+        from instaslice_amd.utils import start_stack_sampler
+
+        os.makedirs(prof_dir, exist_ok=True)
+        prof_stop = start_stack_sampler(
+            os.path.join(prof_dir, f"rank-{rank}.samples"))
+
     # GC latency hygiene for the agent+bench process (control-plane
     # processes tune themselves in runtime/controlplane.py)
     import gc
@@ -428,6 +437,8 @@ def main() -> int:
         print(json.dumps(result), flush=True)
 
     # teardown
+    if prof_stop is not None:
+        prof_stop()
     events.stop()
     agent.stop()
     store.close()

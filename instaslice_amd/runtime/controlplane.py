@@ -45,6 +45,16 @@ def _run_controller_shard(
     from instaslice_amd.controller.reconciler import PodController
     from instaslice_amd.store.netstore import NetStoreClient
 
+    import os
+
+    prof_dir = os.environ.get("INSTASLICE_PROFILE_DIR")
+    prof_stop = None
+    if prof_dir:
+        from instaslice_amd.utils import start_stack_sampler
+
+        os.makedirs(prof_dir, exist_ok=True)
+        prof_stop = start_stack_sampler(
+            os.path.join(prof_dir, f"shard-{shard_index}.samples"))
     store = NetStoreClient("127.0.0.1", port)
     controller = PodController(
         store,
@@ -63,6 +73,8 @@ def _run_controller_shard(
         pass
     controller.stop()
     store.close()
+    if prof_stop is not None:
+        prof_stop()
 
 
 def run_control_plane(

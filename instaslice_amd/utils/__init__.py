@@ -40,3 +40,47 @@ def now() -> float:
 
 def monotonic_ms() -> float:
     return time.monotonic() * 1000.0
+
+
+def start_stack_sampler(path: str, interval_s: float = 0.005):
+    """Wall-clock sampling profiler over ALL threads (sys._current_frames):
+    cProfile only sees the calling thread, but the engines do their work on
+    pump/worker threads. Returns a stop() that writes aggregated
+    `count<TAB>thread<TAB>func@file:line` lines, innermost frame only —
+    enough to see where time (including blocking) goes. Env-gated via
+    INSTASLICE_PROFILE_DIR in the control plane and bench."""
+    import collections
+    import sys
+    import threading
+    import time as _time
+
+    counts = collections.Counter()
+    stop_flag = threading.Event()
+
+    def sample():
+        me = threading.get_ident()
+        names = {}
+        while not stop_flag.is_set():
+            for t in threading.enumerate():
+                names[t.ident] = t.name
+            for tid, frame in sys._current_frames().items():
+                if tid == me:
+                    continue
+                co = frame.f_code
+                key = (names.get(tid, str(tid)),
+                       f"{co.co_name}@{co.co_filename.rsplit('/', 1)[-1]}"
+                       f":{frame.f_lineno}")
+                counts[key] += 1
+            _time.sleep(interval_s)
+
+    t = threading.Thread(target=sample, daemon=True, name="stack-sampler")
+    t.start()
+
+    def stop():
+        stop_flag.set()
+        t.join(timeout=2.0)
+        with open(path, "w") as f:
+            for (tname, loc), n in counts.most_common():
+                f.write(f"{n}\t{tname}\t{loc}\n")
+
+    return stop

@@ -139,25 +139,31 @@ class Engine:
 
     # -- pumps & workers ----------------------------------------------------
 
+    def _handle_event(self, spec: WatchSpec, event_type: str, obj: dict) -> None:
+        """Cache + enqueue for one event. Runs either on a pump thread
+        (queue-mode watches) or DIRECTLY on the store's notify/reader thread
+        (push-mode set_callback) — one fewer handoff per event; must stay
+        fast and never call the store."""
+        md = obj.get("metadata", {})
+        own_key = (obj["kind"], md.get("namespace", ""), md["name"])
+        with self._cache_lock:
+            if event_type == "DELETED":
+                self._cache.pop(own_key, None)
+            else:
+                self._cache[own_key] = obj
+        if spec.map_fn is None:
+            keys = [own_key]
+        else:
+            keys = spec.map_fn(event_type, obj)
+        for k in keys:
+            self.enqueue(k)
+
     def _pump(self, spec: WatchSpec, watch: Watch) -> None:
         while not self._stopping:
             ev = watch.next(timeout=0.5)
             if ev is None:
                 continue
-            event_type, obj = ev
-            md = obj.get("metadata", {})
-            own_key = (obj["kind"], md.get("namespace", ""), md["name"])
-            with self._cache_lock:
-                if event_type == "DELETED":
-                    self._cache.pop(own_key, None)
-                else:
-                    self._cache[own_key] = obj
-            if spec.map_fn is None:
-                keys = [own_key]
-            else:
-                keys = spec.map_fn(event_type, obj)
-            for k in keys:
-                self.enqueue(k)
+            self._handle_event(spec, ev[0], ev[1])
 
     def cached(self, key: Key) -> Optional[dict]:
         """Latest watched object for `key`, or None. READ-ONLY — watch event
@@ -225,12 +231,22 @@ class Engine:
         for spec in self.watch_specs:
             w = self.store.watch(spec.kind, replay=True, filters=spec.filters)
             self._watches.append(w)
-            t = threading.Thread(
-                target=self._pump, args=(spec, w), name=f"{self.name}-pump-{spec.kind}",
-                daemon=True,
-            )
-            t.start()
-            self._threads.append(t)
+            if hasattr(w, "set_callback"):
+                # push mode: events invoke cache+enqueue straight from the
+                # notify/reader thread — no pump thread, no queue handoff
+                def make_cb(sp):
+                    def cb(event_type, obj, _sp=sp):
+                        self._handle_event(_sp, event_type, obj)
+                    return cb
+
+                w.set_callback(make_cb(spec))
+            else:
+                t = threading.Thread(
+                    target=self._pump, args=(spec, w),
+                    name=f"{self.name}-pump-{spec.kind}", daemon=True,
+                )
+                t.start()
+                self._threads.append(t)
         for i in range(self.workers):
             t = threading.Thread(target=self._work, name=f"{self.name}-worker-{i}", daemon=True)
             t.start()

@@ -90,6 +90,40 @@ struct Conn {
   std::deque<OutItem> outbox;
   bool closing = false;
   std::vector<std::shared_ptr<WatchSub>> subs;
+  // serializes actual socket writes between the writer thread and the
+  // reader's inline response path (send_now) — byte interleave guard
+  std::mutex write_mu;
+
+  bool raw_send(const std::string& data) {
+    std::lock_guard<std::mutex> g(write_mu);
+    const char* p = data.data();
+    size_t left = data.size();
+    while (left > 0) {
+      ssize_t n = ::send(fd, p, left, MSG_NOSIGNAL);
+      if (n <= 0) return false;
+      p += n;
+      left -= static_cast<size_t>(n);
+    }
+    return true;
+  }
+
+  // responses skip the outbox + writer wakeup when nothing is queued —
+  // one fewer thread handoff on every request's critical path. A response
+  // may overtake QUEUED watch events; the protocol tolerates that
+  // (responses and events resolve independently client-side; only
+  // event-vs-event order matters, and that stays in the outbox).
+  void send_now(const std::string& frame_bytes) {
+    {
+      std::lock_guard<std::mutex> g(out_mu);
+      if (closing) return;
+    }
+    if (!raw_send(frame_bytes)) {
+      std::lock_guard<std::mutex> g(out_mu);
+      closing = true;
+      outbox.clear();
+      out_cv.notify_one();
+    }
+  }
 
   void enqueue(std::string frame) {
     {
@@ -671,18 +705,11 @@ void writer_loop(std::shared_ptr<Conn> conn) {
     std::string data;
     for (auto& it : batch)
       data += (it.wid >= 0) ? pack_event(it) : std::move(it.bytes);
-    const char* p = data.data();
-    size_t left = data.size();
-    while (left > 0) {
-      ssize_t n = ::send(conn->fd, p, left, MSG_NOSIGNAL);
-      if (n <= 0) {
-        std::lock_guard<std::mutex> g(conn->out_mu);
-        conn->closing = true;
-        conn->outbox.clear();
-        return;
-      }
-      p += n;
-      left -= static_cast<size_t>(n);
+    if (!conn->raw_send(data)) {
+      std::lock_guard<std::mutex> g(conn->out_mu);
+      conn->closing = true;
+      conn->outbox.clear();
+      return;
     }
   }
 }
@@ -739,13 +766,13 @@ void reader_loop(Store& store, std::shared_ptr<Conn> conn) {
         }
         const Value* replayv = req.find("replay");
         bool replay = !replayv || replayv->truthy();
-        // respond first so the watch_id precedes any event on the wire
-        // (both land in the same ordered outbox)
+        // respond first so the watch_id precedes any replay event on the
+        // wire: inline send happens-before add_watch enqueues the replay
         Value result = Value::map();
         result.setkey("watch_id", Value::integer(sub->watch_id));
         std::string payload;
         pack(ok_response(rid, std::move(result)), payload);
-        conn->enqueue(frame(payload));
+        conn->send_now(frame(payload));
         conn->subs.push_back(sub);
         store.add_watch(sub, replay);
         continue;
@@ -783,7 +810,7 @@ void reader_loop(Store& store, std::shared_ptr<Conn> conn) {
     }
     std::string payload;
     pack(resp, payload);
-    conn->enqueue(frame(payload));
+    conn->send_now(frame(payload));
   }
 
   // teardown

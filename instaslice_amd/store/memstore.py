@@ -156,15 +156,45 @@ class Watch:
         self.filters = filters
         self._q: "queue.Queue[Optional[Tuple[str, dict]]]" = queue.Queue()
         self._stopped = False
+        self._cb = None
+        self._cb_lock = threading.Lock()
 
     def _matches(self, obj: dict) -> bool:
         if self.filters is not None:
             return any(_filter_matches(f, obj) for f in self.filters)
         return self.kind is None or self.kind == obj["kind"]
 
+    def set_callback(self, fn) -> None:
+        """Push-mode delivery: events invoke `fn(event_type, obj)` directly
+        on the notifying thread instead of landing in the queue — one fewer
+        thread handoff per event for consumers that only need a cheap
+        enqueue (the reconcile engines). `fn` MUST be fast and non-blocking
+        and MUST NOT call back into the store. Queued backlog (e.g. the
+        replay burst between watch() and set_callback) is drained into `fn`
+        first, in order."""
+        with self._cb_lock:
+            while True:
+                try:
+                    ev = self._q.get_nowait()
+                except queue.Empty:
+                    break
+                if ev is not None:
+                    fn(*ev)
+            self._cb = fn
+
     def _push(self, event: Tuple[str, dict]) -> None:
-        if not self._stopped:
-            self._q.put(event)
+        if self._stopped:
+            return
+        with self._cb_lock:
+            if self._cb is not None:
+                try:
+                    self._cb(*event)
+                except Exception:  # noqa: BLE001 - consumer bug must not kill notify
+                    import traceback
+
+                    traceback.print_exc()
+                return
+        self._q.put(event)
 
     def next(self, timeout: Optional[float] = None) -> Optional[Tuple[str, dict]]:
         """Blocking next event; None on stop or timeout."""

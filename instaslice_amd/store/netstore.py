@@ -196,12 +196,40 @@ class StoreServer:
 
 
 class _ClientWatch:
-    """Client-side watch mirroring memstore.Watch."""
+    """Client-side watch mirroring memstore.Watch (incl. push-mode
+    set_callback — see memstore.Watch.set_callback for the contract)."""
 
     def __init__(self) -> None:
         self._q: "queue.Queue[Optional[Tuple[str, dict]]]" = queue.Queue()
         self._stopped = False
         self.kind: Optional[str] = None
+        self._cb = None
+        self._cb_lock = threading.Lock()
+
+    def set_callback(self, fn) -> None:
+        with self._cb_lock:
+            while True:
+                try:
+                    ev = self._q.get_nowait()
+                except queue.Empty:
+                    break
+                if ev is not None:
+                    fn(*ev)
+            self._cb = fn
+
+    def _deliver(self, event) -> None:
+        if self._stopped:
+            return
+        with self._cb_lock:
+            if self._cb is not None:
+                try:
+                    self._cb(*event)
+                except Exception:  # noqa: BLE001 - consumer bug must not kill reader
+                    import traceback
+
+                    traceback.print_exc()
+                return
+        self._q.put(event)
 
     def next(self, timeout: Optional[float] = None) -> Optional[Tuple[str, dict]]:
         try:
@@ -260,8 +288,8 @@ class NetStoreClient:
                 w = self._watches.get(wid)
                 if w is None:
                     self._orphan_events.setdefault(wid, []).append((et, obj))
-            if w is not None and not w._stopped:
-                w._q.put((et, obj))
+            if w is not None:
+                w._deliver((et, obj))
             return
         ev = self._pending.pop(msg.get("id"), None)
         if ev is not None:
@@ -429,7 +457,7 @@ class NetStoreClient:
             orphans = self._orphan_events.pop(wid, [])
         # deliver events that raced ahead of this registration, in order
         for ev in orphans:
-            w._q.put(ev)
+            w._deliver(ev)
         return w
 
     def update_with_retry(

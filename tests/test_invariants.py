@@ -4,6 +4,7 @@ controller+agent stack on FakeAmdSmi; after settling, the CR must satisfy the
 structural invariants the whole design rests on (no double-booked ordinal,
 mode coherence, prepared/allocation/ConfigMap/capacity 1:1)."""
 
+import os
 import time
 
 import pytest
@@ -108,7 +109,9 @@ def _check_invariants(c: Cluster):
 
 
 @pytest.mark.parametrize("policy", ["packed-fit", "first-fit", "spread-fit"])
-@settings(max_examples=10, deadline=None,
+@settings(max_examples=int(os.environ.get("INSTASLICE_INVARIANT_EXAMPLES",
+                                           "10")),
+          deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(ops=ops_strategy)
 def test_random_lifecycle_invariants(policy, ops):

@@ -12,8 +12,10 @@ One step = one full pod lifecycle on this rank's node:
   (latency sample) -> HIP payload kernel runs inside the allocated partition
   -> pod deleted -> allocation drained.
 
-Topology: rank 0 hosts the store (TCP) + the cluster controller; every rank
-(including 0) runs one node agent managing its own GPU. The bench process
+Topology: rank 0 spawns the control-plane PROCESS (the native store daemon
+instaslice-stored when built, plus the controller — sharded across processes
+when the machine has idle cores); every rank (including 0) runs one node
+agent managing its own GPU over TCP. The bench process
 NEVER opens a HIP context of its own: amdgpu refuses partition mode changes
 while any process holds the GPU, so payload kernels run in short-lived child
 processes with the pod's ROCR_VISIBLE_DEVICES — exactly like real pods — and

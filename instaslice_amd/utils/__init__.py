@@ -12,19 +12,46 @@ _LOG_CONFIGURED = False
 _LOCK = threading.Lock()
 
 
-def get_logger(name: str) -> logging.Logger:
-    """Structured-ish logger (the reference uses zap/logr; we use stdlib).
+class _JsonFormatter(logging.Formatter):
+    """One JSON object per line (the reference's zap production encoder
+    analog) — for log aggregators. Enabled with INSTASLICE_LOG_JSON=1."""
 
-    Level comes from INSTASLICE_LOG_LEVEL (default INFO).
+    def format(self, record: logging.LogRecord) -> str:
+        import json
+
+        out = {
+            "ts": round(record.created, 3),
+            "level": record.levelname,
+            "logger": record.name,
+            "msg": record.getMessage(),
+        }
+        if record.exc_info:
+            out["exc"] = self.formatException(record.exc_info)
+        return json.dumps(out)
+
+
+def get_logger(name: str) -> logging.Logger:
+    """Structured logger (the reference uses zap/logr; we use stdlib).
+
+    Level from INSTASLICE_LOG_LEVEL (default INFO); INSTASLICE_LOG_JSON=1
+    switches to one-JSON-object-per-line output.
     """
     global _LOG_CONFIGURED
     with _LOCK:
         if not _LOG_CONFIGURED:
             level = os.environ.get("INSTASLICE_LOG_LEVEL", "INFO").upper()
-            logging.basicConfig(
-                level=getattr(logging, level, logging.INFO),
-                format="%(asctime)s %(levelname)-5s %(name)s %(message)s",
-            )
+            if os.environ.get("INSTASLICE_LOG_JSON", "") not in ("", "0"):
+                handler = logging.StreamHandler()
+                handler.setFormatter(_JsonFormatter())
+                logging.basicConfig(
+                    level=getattr(logging, level, logging.INFO),
+                    handlers=[handler],
+                )
+            else:
+                logging.basicConfig(
+                    level=getattr(logging, level, logging.INFO),
+                    format="%(asctime)s %(levelname)-5s %(name)s %(message)s",
+                )
             _LOG_CONFIGURED = True
     return logging.getLogger(name)
 

@@ -26,6 +26,7 @@ import time
 from typing import Dict, List, Optional
 
 from instaslice_amd import POD_RESOURCE_PREFIX
+from instaslice_amd.api.events import emit
 from instaslice_amd.api.types import (
     AllocationStatus,
     GpuStatus,
@@ -341,8 +342,6 @@ class NodeAgent:
             # that change re-reconciles us with fresh state
             self.log.debug("prepare commit superseded: %s", commit["error"]["msg"])
             return False
-        from instaslice_amd.api.events import emit
-
         emit(self.store,
              {"kind": "Pod", "namespace": alloc["namespace"],
               "name": alloc["podName"]},
@@ -432,8 +431,6 @@ class NodeAgent:
                         alloc["podName"], gpu[:8], e,
                     )
                     self._mode_set_failures.pop(gpu, None)
-                    from instaslice_amd.api.events import emit
-
                     emit(self.store,
                          {"kind": "Pod", "namespace": alloc["namespace"],
                           "name": alloc["podName"]},
@@ -500,8 +497,6 @@ class NodeAgent:
             from instaslice_amd.metrics import get_metrics
 
             get_metrics().allocation("deleted")
-            from instaslice_amd.api.events import emit
-
             emit(self.store,
                  {"kind": "Pod", "namespace": alloc["namespace"],
                   "name": alloc["podName"]},

@@ -24,8 +24,10 @@ Differences by design (not omissions):
 from __future__ import annotations
 
 import time
+import zlib
 from typing import List, Optional, Tuple
 
+from instaslice_amd.api.events import emit
 from instaslice_amd.api.types import (
     AllocationDetails,
     AllocationStatus,
@@ -116,8 +118,6 @@ class PodController:
     def _owns(self, namespace: str, name: str) -> bool:
         if self.shard_count <= 1:
             return True
-        import zlib
-
         h = zlib.crc32(f"{namespace}/{name}".encode())
         return h % self.shard_count == self.shard_index
 
@@ -258,8 +258,6 @@ class PodController:
                     md["name"], profile_name, node_name,
                     placement.gpu_uuid[:8], placement.ordinal,
                 )
-                from instaslice_amd.api.events import emit
-
                 emit(self.store, pod, "Placed",
                      f"allocated {profile_name} on {node_name}/"
                      f"{placement.gpu_uuid[:8]}#{placement.ordinal}"
@@ -271,8 +269,6 @@ class PodController:
                 return Result(
                     requeue_after=self.node_stale_after_s or None
                 )
-        from instaslice_amd.api.events import emit
-
         emit(self.store, pod, "Unschedulable",
              f"no node can host profile {profile_name} right now",
              type_="Warning")
@@ -389,8 +385,6 @@ class PodController:
                                               AllocationStatus.UNGATED)
         )
         if have < want:
-            from instaslice_amd.api.events import emit
-
             emit(self.store, pod, "GangWaiting",
                  f"partition ready; waiting for gang '{group}' "
                  f"({have}/{want} realized)")
@@ -455,8 +449,6 @@ class PodController:
             ], quiet=True)
         except NotFound:
             pass
-        from instaslice_amd.api.events import emit
-
         emit(self.store, pod, "Preempting",
              f"evicting {v_alloc['namespace']}/{v_alloc['podName']} "
              f"(priority {vprio}) and nominating its slot", type_="Warning")

@@ -539,12 +539,26 @@ class NodeAgent:
 
     def _heartbeat_loop(self) -> None:
         while not self._hb_stop.is_set():
+            # liveness + GPU health telemetry in one patch: operators (and
+            # the controller's staleness check) read status.heartbeat;
+            # status.gpuMetrics carries the live amd-smi counters per GPU —
+            # the north star's "amd-smi counters captured" made continuous,
+            # not just per-reconfigure
+            metrics = {}
+            for uuid in list(self._gpus):
+                try:
+                    m = self.smi.get_metrics(uuid)
+                except (SmiError, SmiNotSupported):
+                    continue
+                metrics[uuid] = {k: round(float(v), 3) for k, v in m.items()}
+            ops = [{"op": "set", "path": ["status", "heartbeat"],
+                    "value": time.time()}]
+            if metrics:
+                ops.append({"op": "set", "path": ["status", "gpuMetrics"],
+                            "value": metrics})
             try:
-                self.store.patch(
-                    "Instaslice", self.node_name, INSTASLICE_NS,
-                    [{"op": "set", "path": ["status", "heartbeat"],
-                      "value": time.time()}], quiet=True,
-                )
+                self.store.patch("Instaslice", self.node_name, INSTASLICE_NS,
+                                 ops, quiet=True)
             except Exception as e:  # store outage: keep trying
                 self.log.warning("heartbeat failed: %s", e)
             self._hb_stop.wait(self.heartbeat_every_s)

@@ -150,7 +150,18 @@ def crd_manifest() -> dict:
                                 "spec": SPEC_SCHEMA,
                                 "status": {
                                     "type": "object",
-                                    "properties": {"processed": _STR},
+                                    "properties": {
+                                        "processed": _STR,
+                                        "heartbeat": {"type": "number"},
+                                        "gpuMetrics": {
+                                            "type": "object",
+                                            "additionalProperties": {
+                                                "type": "object",
+                                                "additionalProperties": {
+                                                    "type": "number"},
+                                            },
+                                        },
+                                    },
                                 },
                             },
                         }

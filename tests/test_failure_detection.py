@@ -110,3 +110,31 @@ def test_creating_allocation_reclaimed_from_dead_node():
             healthy.stop()
     finally:
         controller.stop()
+
+
+def test_heartbeat_carries_gpu_metrics():
+    """status.gpuMetrics: live amd-smi counters ride every heartbeat (the
+    north-star observability requirement made continuous)."""
+    import time
+
+    from instaslice_amd.agent.daemonset import NodeAgent
+    from instaslice_amd.smi.fake import FakeAmdSmi
+    from instaslice_amd.store.memstore import MemStore
+
+    store = MemStore()
+    agent = NodeAgent(store, FakeAmdSmi(num_gpus=2, node_name="n0"), "n0",
+                      heartbeat_every_s=0.1)
+    agent.start()
+    try:
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            cr = store.get("Instaslice", "n0", "instaslice-system")
+            gm = (cr.get("status") or {}).get("gpuMetrics") or {}
+            if len(gm) == 2:
+                break
+            time.sleep(0.05)
+        assert len(gm) == 2
+        for m in gm.values():
+            assert "gfx_activity_pct" in m
+    finally:
+        agent.stop()

@@ -152,11 +152,23 @@ struct Conn {
 
   // push without waking the writer: callers holding the store mutex defer
   // the futex wake until after they release it (notify_one is a syscall;
-  // at ~200k event pushes/s it is measurable critical-section time)
+  // at ~200k event pushes/s it is measurable critical-section time).
+  // Slow-consumer guard: a client that stops reading would grow its outbox
+  // without bound — past the cap the connection is closed (the k8s API
+  // server likewise terminates watchers that can't keep up; a
+  // reconnect-enabled client resyncs with replay).
+  static constexpr size_t kMaxOutbox = 65536;
+
   void enqueue_event_silent(int64_t wid, const char* type,
                             std::shared_ptr<const Value> obj) {
     std::lock_guard<std::mutex> g(out_mu);
     if (closing) return;
+    if (outbox.size() >= kMaxOutbox) {
+      closing = true;
+      outbox.clear();
+      ::shutdown(fd, SHUT_RDWR);  // unblocks reader+writer; conn tears down
+      return;
+    }
     OutItem it;
     it.wid = wid;
     it.ev_type = type;

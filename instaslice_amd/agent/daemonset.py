@@ -25,7 +25,7 @@ import threading
 import time
 from typing import Dict, List, Optional
 
-from instaslice_amd import POD_RESOURCE_PREFIX
+from instaslice_amd import POD_RESOURCE_PREFIX, RESOURCE_PREFIX
 from instaslice_amd.api.events import emit
 from instaslice_amd.api.types import (
     AllocationStatus,
@@ -85,6 +85,7 @@ class NodeAgent:
         self.heartbeat_every_s = heartbeat_every_s
         self._hb_stop = threading.Event()
         self._hb_thread = None
+        self._last_capacity: Dict[str, int] = {}
         # filtered watch: only THIS node's CR crosses the wire (a cluster
         # with N agents otherwise broadcasts every CR event to all N)
         self.engine = Engine(
@@ -537,6 +538,32 @@ class NodeAgent:
 
     # -- lifecycle ------------------------------------------------------------
 
+    def _profile_capacity(self, cr: dict) -> Dict[str, int]:
+        """Free slots per profile on this node right now: free ordinals on
+        GPUs already in the profile's mode + idle flippable GPUs times the
+        profile's partition count. This is what the AMD device plugin would
+        advertise as extended resources; published on the heartbeat for
+        schedulers/autoscalers (eventual consistency is fine there)."""
+        from instaslice_amd.controller.policy import build_gpu_views
+        from instaslice_amd.partition.profiles import ProfileCatalog
+
+        spec = cr.get("spec", {})
+        placements = spec.get("placements") or {}
+        if not placements:
+            return {}
+        cat = ProfileCatalog.from_dict(placements)
+        views = build_gpu_views(self.node_name, spec)
+        out: Dict[str, int] = {}
+        for prof in cat.profiles:
+            n = 0
+            for v in views:
+                if v.compute_mode is prof.compute:
+                    n += max(0, prof.partitions_per_gpu - len(v.occupied))
+                elif not v.occupied and not v.mode_locked:
+                    n += prof.partitions_per_gpu
+            out[prof.name] = n
+        return out
+
     def _heartbeat_loop(self) -> None:
         while not self._hb_stop.is_set():
             # liveness + GPU health telemetry in one patch: operators (and
@@ -556,9 +583,26 @@ class NodeAgent:
             if metrics:
                 ops.append({"op": "set", "path": ["status", "gpuMetrics"],
                             "value": metrics})
+            reqs = [{"verb": "patch", "kind": "Instaslice",
+                     "name": self.node_name, "namespace": INSTASLICE_NS,
+                     "ops": ops}]
+            cr = self.engine.cached(
+                ("Instaslice", INSTASLICE_NS, self.node_name))
+            if cr is not None:
+                cap = self._profile_capacity(cr)
+                if cap and cap != self._last_capacity:
+                    self._last_capacity = cap
+                    reqs.append(
+                        {"verb": "patch", "kind": "Node",
+                         "name": self.node_name, "namespace": "", "ops": [
+                             {"op": "set",
+                              "path": ["status", "capacity",
+                                       RESOURCE_PREFIX + prof],
+                              "value": n}
+                             for prof, n in sorted(cap.items())
+                         ]})
             try:
-                self.store.patch("Instaslice", self.node_name, INSTASLICE_NS,
-                                 ops, quiet=True)
+                self.store.batch(reqs, quiet=True)
             except Exception as e:  # store outage: keep trying
                 self.log.warning("heartbeat failed: %s", e)
             self._hb_stop.wait(self.heartbeat_every_s)

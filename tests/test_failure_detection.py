@@ -138,3 +138,34 @@ def test_heartbeat_carries_gpu_metrics():
             assert "gfx_activity_pct" in m
     finally:
         agent.stop()
+
+
+def test_heartbeat_advertises_profile_capacity():
+    """Node.status.capacity gets amd.com/<profile> free-slot counts on the
+    heartbeat (device-plugin-style extended-resource advertisement for
+    schedulers/autoscalers)."""
+    import time
+
+    from instaslice_amd.agent.daemonset import NodeAgent
+    from instaslice_amd.smi.fake import FakeAmdSmi
+    from instaslice_amd.store.memstore import MemStore
+
+    store = MemStore()
+    agent = NodeAgent(store, FakeAmdSmi(num_gpus=2, node_name="n0"), "n0",
+                      heartbeat_every_s=0.1)
+    agent.start()
+    try:
+        deadline = time.monotonic() + 5
+        cap = {}
+        while time.monotonic() < deadline:
+            node = store.get("Node", "n0", "")
+            cap = {k: v for k, v in (node["status"].get("capacity") or {}).items()
+                   if k.startswith("amd.com/")}
+            if cap:
+                break
+            time.sleep(0.05)
+        # 2 idle SPX GPUs: cpx 16 slots, qpx 8, dpx 4, spx 2
+        assert cap.get("amd.com/cpx-1x36") == 16, cap
+        assert cap.get("amd.com/spx-8x288") == 2, cap
+    finally:
+        agent.stop()

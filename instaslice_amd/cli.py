@@ -239,6 +239,19 @@ def cmd_status(args) -> int:
     return 0
 
 
+def cmd_cordon(args) -> int:
+    """Mark a node unschedulable for NEW placements (running partitions are
+    untouched) — kubectl-cordon analog; `uncordon` reverses it."""
+    store = _connect(args.store)
+    want = not args.uncordon
+    store.patch("Instaslice", args.node, "instaslice-system", [
+        {"op": "set", "path": ["spec", "cordoned"], "value": want},
+    ], quiet=True)
+    print(f"node {args.node} {'cordoned' if want else 'uncordoned'}")
+    store.close()
+    return 0
+
+
 def cmd_describe(args) -> int:
     """Pod detail + its Events (kubectl-describe analog)."""
     store = _connect(args.store)
@@ -398,6 +411,12 @@ def main(argv=None) -> int:
     p = sub.add_parser("status", help="cluster state summary")
     p.add_argument("--store", default="127.0.0.1:7080")
     p.set_defaults(fn=cmd_status)
+
+    p = sub.add_parser("cordon", help="drain a node for new placements")
+    p.add_argument("--store", default="127.0.0.1:7080")
+    p.add_argument("--node", required=True)
+    p.add_argument("--uncordon", action="store_true")
+    p.set_defaults(fn=cmd_cordon)
 
     p = sub.add_parser("describe", help="pod detail + events")
     p.add_argument("--store", default="127.0.0.1:7080")

@@ -199,6 +199,8 @@ class PodController:
                 continue
             if self._node_stale(cr):
                 continue  # agent dead: no new placements here
+            if (cr.get("spec") or {}).get("cordoned"):
+                continue  # operator drained this node (CLI cordon)
             profile = self._profile_for(cr, profile_name)
             if profile is None:
                 continue  # node does not offer this profile

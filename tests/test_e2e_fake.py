@@ -526,3 +526,27 @@ def test_gang_ungate_barrier():
             c.wait_pod_scheduled(n, timeout=10.0)
     finally:
         c.stop()
+
+
+def test_cordon_drains_new_placements():
+    """spec.cordoned: no new placements on the node; running pods stay;
+    uncordon restores placement (kubectl-cordon analog)."""
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        c.submit_pod("before", "cpx-1x36")
+        c.wait_pod_scheduled("before")
+        c.store.patch("Instaslice", "node-0", "instaslice-system", [
+            {"op": "set", "path": ["spec", "cordoned"], "value": True},
+        ])
+        c.submit_pod("during", "cpx-1x36")
+        assert c.wait_pod_outcome("during") == "unschedulable"
+        assert "before" in {a["podName"]
+                            for a in c.allocations("node-0").values()}
+        c.store.patch("Instaslice", "node-0", "instaslice-system", [
+            {"op": "set", "path": ["spec", "cordoned"], "value": False},
+        ])
+        c.wait_pod_scheduled("during", timeout=10.0)
+    finally:
+        c.stop()

@@ -63,9 +63,55 @@ ROCR_VISIBLE_DEVICES=$DEV timeout 60 ./instaslice_amd/bin/instaslice-payload inf
 log "curl the agent metrics endpoint"
 timeout 10 curl -s http://127.0.0.1:18084/metrics | grep -E "instaslice_(alloc|partition|reconcile)" | head -12 >> $OUT 2>&1
 
+log "python -m instaslice_amd describe --name demo   (events timeline)"
+timeout 30 python -m instaslice_amd describe --store 127.0.0.1:7080 --name demo >> $OUT 2>&1
+
 log "python -m instaslice_amd delete --name demo"
 timeout 30 python -m instaslice_amd delete --store 127.0.0.1:7080 --name demo >> $OUT 2>&1
 sleep 2
+
+log "PREEMPTION DEMO: fill the node at priority 1, then submit priority 10"
+timeout 60 python - <<PYEOF >> $OUT 2>&1
+import time
+from instaslice_amd.store.netstore import NetStoreClient
+from instaslice_amd.api.types import new_pod
+c = NetStoreClient("127.0.0.1", 7080)
+c.create(new_pod("low", profile="$PROFILE", priority=1))
+deadline = time.monotonic() + 20
+while time.monotonic() < deadline:
+    if not c.get("Pod", "low", "default")["spec"].get("schedulingGates"):
+        break
+    time.sleep(0.05)
+print("low-priority pod holds the partition")
+t0 = time.perf_counter()
+c.create(new_pod("high", profile="$PROFILE", priority=10))
+while time.monotonic() < deadline:
+    try:
+        if not c.get("Pod", "high", "default")["spec"].get("schedulingGates"):
+            break
+    except Exception:
+        pass
+    time.sleep(0.01)
+print("high-priority pod scheduled in %.1f ms (low evicted)"
+      % ((time.perf_counter() - t0) * 1e3))
+for name in ("high.Preempting", "low.Preempted"):
+    ev = c.get("Event", name, "default")
+    print(f"event {name}: {ev['message']}")
+c.delete("Pod", "high", "default")
+c.close()
+PYEOF
+
+log "CORDON DEMO: drain the node, submit, uncordon"
+timeout 30 python -m instaslice_amd cordon --store 127.0.0.1:7080 --node mi355x-node >> $OUT 2>&1
+timeout 30 python -m instaslice_amd submit --store 127.0.0.1:7080 --name held --profile $PROFILE >> $OUT 2>&1
+sleep 1
+timeout 30 python -m instaslice_amd describe --store 127.0.0.1:7080 --name held | grep -A3 Unschedulable >> $OUT 2>&1
+timeout 30 python -m instaslice_amd cordon --store 127.0.0.1:7080 --node mi355x-node --uncordon >> $OUT 2>&1
+timeout 60 python -m instaslice_amd submit --store 127.0.0.1:7080 --name held2 --profile $PROFILE --wait --timeout 30 > /dev/null 2>&1
+echo "uncordoned; placements resumed" >> $OUT
+timeout 30 python -m instaslice_amd delete --store 127.0.0.1:7080 --name held >> $OUT 2>&1
+timeout 30 python -m instaslice_amd delete --store 127.0.0.1:7080 --name held2 >> $OUT 2>&1
+sleep 1
 
 log "python -m instaslice_amd status   (drained)"
 timeout 30 python -m instaslice_amd status --store 127.0.0.1:7080 >> $OUT 2>&1

@@ -393,11 +393,16 @@ def main() -> int:
     sync()
     elapsed = time.perf_counter() - t_start
 
+    import resource
+
+    rss_mb = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024.0
+
     # max over ranks (the contract), pooled latency samples
     if world > 1:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        t = torch.tensor([elapsed, rss_mb], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
+        elapsed = float(t[0].item())
+        rss_mb = float(t[1].item())
         pooled = [None] * world
         dist.all_gather_object(pooled, latencies_ms)
         all_lat = [x for sub in pooled for x in sub]
@@ -424,6 +429,7 @@ def main() -> int:
             "vs_baseline": None,  # reference publishes no numbers
             "dtype": "fp32",
             "data": "synthetic",
+            "max_rank_rss_mb": round(rss_mb, 1),
             "p50_alloc_latency_ms": round(p50, 3) if p50 is not None else None,
             "p99_alloc_latency_ms": round(p99, 3) if p99 is not None else None,
             "config": {

@@ -214,11 +214,18 @@ def cmd_status(args) -> int:
         spec = cr.get("spec", {})
         out["nodes"].append({
             "node": cr["metadata"]["name"],
+            "cordoned": bool(spec.get("cordoned")),
+            "heartbeatAgeS": (
+                round(time.time() - float(cr["status"]["heartbeat"]), 1)
+                if (cr.get("status") or {}).get("heartbeat") else None
+            ),
             "gpus": {
                 u[:8]: {
                     "mode": f"{g.get('computeMode')}/{g.get('memoryMode')}",
                     "used": g.get("usedOrdinals", []),
                     "modeLocked": g.get("modeLocked", False),
+                    "metrics": ((cr.get("status") or {})
+                                .get("gpuMetrics", {}).get(u)),
                 }
                 for u, g in (spec.get("gpus") or {}).items()
             },

@@ -8,7 +8,7 @@ ROCM_ARCH ?= gfx950
 
 all: native
 
-native:  ## compile partitiond + payload for $(ROCM_ARCH), in-tree
+native:  ## compile partitiond + payload + instaslice-stored, in-tree
 	PYTORCH_ROCM_ARCH=$(ROCM_ARCH) $(PYTHON) build_native.py
 
 test:    ## CPU test suite (no GPU needed)
@@ -18,7 +18,7 @@ test-gpu:  ## GPU tier (run on a MI355X box)
 	$(PYTHON) -m pytest tests/ -q -m gpu
 
 bench:   ## flagship benchmark, single GPU
-	$(PYTHON) bench.py --steps 100 --warmup 10
+	$(PYTHON) bench.py
 
 scenarios:  ## BASELINE.json measurement configs 1-5 (fake-SMI)
 	$(PYTHON) -m benchmarks.scenarios

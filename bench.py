@@ -35,6 +35,11 @@ from __future__ import annotations
 import argparse
 import json
 import os
+
+# quiet by default: a multi-minute bench logs one INFO per reconfigure per
+# pod in live mode — megabytes of noise at 3000+ pods/s (set
+# INSTASLICE_LOG_LEVEL explicitly to override)
+os.environ.setdefault("INSTASLICE_LOG_LEVEL", "WARNING")
 import random
 import statistics
 import sys

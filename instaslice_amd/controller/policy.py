@@ -103,7 +103,12 @@ def build_gpu_views(node_name: str, spec: dict,
             continue
         if now - float(nom.get("ts", 0)) > NOMINATION_TTL_S:
             continue
-        occupied.setdefault(nom["gpuUUID"], set()).add(nom["ordinal"])
+        if nom.get("wholeGpu"):
+            # whole-GPU reservation (multi-victim preemption): nobody else
+            # may take any ordinal or flip it while the nominee waits
+            occupied.setdefault(nom["gpuUUID"], set()).update(range(8))
+        else:
+            occupied.setdefault(nom["gpuUUID"], set()).add(nom["ordinal"])
 
     views: List[GpuView] = []
     for uuid, gd in (spec.get("gpus") or {}).items():

@@ -405,11 +405,14 @@ class PodController:
 
     def _maybe_preempt(self, pod: dict, profile_name: str,
                        crs: List[dict]) -> bool:
-        """Evict ONE lowest-priority, strictly-lower-priority allocation of
-        the SAME profile (exact slot fit; cross-profile/mode-flip preemption
-        is intentionally out of scope — it would need multi-victim planning).
-        Returns True if an eviction was issued. A per-pod cooldown prevents
-        cascading evictions while the first victim is still draining."""
+        """Preemption planner, cheapest plan first:
+        1. ONE lowest-priority same-profile victim (exact slot fit);
+        2. else a WHOLE GPU whose every allocation is strictly lower
+           priority — all of them are evicted and the GPU is nominated
+           (mode-flip placement once it drains), which lets e.g. an SPX
+           job displace a handful of low-priority CPX pods.
+        Returns True if evictions were issued. A per-pod cooldown prevents
+        cascading evictions while victims are still draining."""
         prio = self._pod_priority(pod)
         if prio <= 0:
             return False
@@ -433,7 +436,7 @@ class PodController:
                 if victim is None or vprio < victim[0]:
                     victim = (vprio, cr["metadata"]["name"], a)
         if victim is None:
-            return False
+            return self._preempt_whole_gpu(pod, profile_name, prio, uid, crs)
         vprio, v_node, v_alloc = victim
         try:
             self.store.delete("Pod", v_alloc["podName"], v_alloc["namespace"])
@@ -467,6 +470,76 @@ class PodController:
             v_alloc["namespace"], v_alloc["podName"], vprio,
             pod["metadata"]["name"], prio,
             v_alloc["gpuUUID"][:8], v_alloc["ordinal"],
+        )
+        from instaslice_amd.metrics import get_metrics
+
+        get_metrics().allocation("preempted")
+        return True
+
+    def _preempt_whole_gpu(self, pod: dict, profile_name: str, prio: int,
+                           uid: str, crs: List[dict]) -> bool:
+        """Plan 2: evict every allocation on the GPU whose max victim
+        priority is lowest (fewest victims as tie-break) and nominate the
+        WHOLE GPU for this pod — once drained it is idle and flippable to
+        the requested profile's mode."""
+        best = None  # ((max_vprio, n), cr_name, gpu_uuid, victims)
+        for cr in crs:
+            if self._node_stale(cr) or (cr.get("spec") or {}).get("cordoned"):
+                continue
+            profile = self._profile_for(cr, profile_name)
+            if profile is None:
+                continue
+            spec = cr["spec"]
+            by_gpu: dict = {}
+            for a in (spec.get("allocations") or {}).values():
+                by_gpu.setdefault(a["gpuUUID"], []).append(a)
+            for g_uuid, gd in (spec.get("gpus") or {}).items():
+                victims = by_gpu.get(g_uuid)
+                if not victims:
+                    continue  # idle GPUs are normal-placement territory
+                if gd.get("modeLocked") and gd.get(
+                    "computeMode"
+                ) != profile.compute.value:
+                    continue  # cannot flip to the needed mode
+                if any(a.get("allocationStatus") == AllocationStatus.DELETED
+                       for a in victims):
+                    return False  # capacity already draining; just wait
+                vmax = max(int(a.get("priority", 0)) for a in victims)
+                if vmax >= prio:
+                    continue
+                key = (vmax, len(victims))
+                if best is None or key < best[0]:
+                    best = (key, cr["metadata"]["name"], g_uuid, victims)
+        if best is None:
+            return False
+        (vmax, n), cr_name, gpu_uuid, victims = best
+        for a in victims:
+            try:
+                self.store.delete("Pod", a["podName"], a["namespace"])
+            except NotFound:
+                pass
+            emit(self.store,
+                 {"kind": "Pod", "namespace": a["namespace"],
+                  "name": a["podName"]},
+                 "Preempted",
+                 f"evicted (whole-GPU preemption) for higher-priority pod "
+                 f"{pod['metadata']['name']} (priority {prio})",
+                 type_="Warning")
+        try:
+            self.store.patch("Instaslice", cr_name, INSTASLICE_NS, [
+                {"op": "set", "path": ["spec", "nominations", uid],
+                 "value": {"gpuUUID": gpu_uuid, "wholeGpu": True,
+                           "ts": time.time()}},
+            ], quiet=True)
+        except NotFound:
+            pass
+        emit(self.store, pod, "Preempting",
+             f"evicting {n} pod(s) (max priority {vmax}) from GPU "
+             f"{gpu_uuid[:8]} and nominating the whole GPU", type_="Warning")
+        self._preempted_for[uid] = (gpu_uuid, time.monotonic())
+        self.log.warning(
+            "whole-GPU preemption: %d pods off %s for %s (priority %d)",
+            n, gpu_uuid[:8], pod["metadata"]["name"], prio,
         )
         from instaslice_amd.metrics import get_metrics
 

@@ -550,3 +550,41 @@ def test_cordon_drains_new_placements():
         c.wait_pod_scheduled("during", timeout=10.0)
     finally:
         c.stop()
+
+
+def test_whole_gpu_preemption_for_bigger_profile():
+    """An SPX pod with high priority displaces a GPU full of low-priority
+    CPX pods (multi-victim whole-GPU preemption + GPU nomination); a single
+    higher-priority CPX pod on the other GPU shields it."""
+    from instaslice_amd.api.types import new_pod
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=2, node_name="node-0"))
+    c.start()
+    try:
+        # gpu A: 8 low-prio CPX; gpu B: 1 HIGH-prio CPX (shielded)
+        for i in range(8):
+            c.store.create(new_pod(f"lo{i}", profile="cpx-1x36", priority=1))
+            c.wait_pod_scheduled(f"lo{i}")
+        c.store.create(new_pod("shield", profile="cpx-1x36", priority=50))
+        c.wait_pod_scheduled("shield")
+        shield_gpu = next(a["gpuUUID"]
+                          for a in c.allocations("node-0").values()
+                          if a["podName"] == "shield")
+        # SPX needs a whole idle GPU; none exists -> whole-GPU preemption
+        c.store.create(new_pod("big", profile="spx-8x288", priority=10))
+        c.wait_pod_scheduled("big", timeout=20.0)
+        allocs = c.allocations("node-0")
+        big = next(a for a in allocs.values() if a["podName"] == "big")
+        assert big["gpuUUID"] != shield_gpu, "evicted the shielded GPU"
+        names = {a["podName"] for a in allocs.values()}
+        assert "shield" in names
+        assert not any(n.startswith("lo") for n in names), names
+        pods = {p["metadata"]["name"] for p in c.store.list("Pod")}
+        assert "shield" in pods and "big" in pods
+        assert not any(n.startswith("lo") for n in pods)
+        # nomination consumed
+        cr = c.store.get("Instaslice", "node-0", "instaslice-system")
+        assert not (cr["spec"].get("nominations") or {})
+    finally:
+        c.stop()

@@ -431,3 +431,35 @@ def test_chaos_store_restart_mid_churn(tmp_path):
         for c in (cstore, astore, bench):
             c.close()
         server.stop()
+
+
+@needs_stored
+def test_daemon_survives_protocol_garbage():
+    """Malformed frames drop only the offending connection; the daemon keeps
+    serving other clients (fuzz-resistance of the wire protocol)."""
+    import socket as socklib
+    import struct
+
+    server = NativeStoreServer().start()
+    good = NetStoreClient("127.0.0.1", server.port)
+    try:
+        good.create(_obj("keep"))
+        for garbage in (
+            b"\x00\x00\x00\x05\xc1\xc1\xc1\xc1\xc1",  # invalid msgpack tag
+            b"\xff\xff\xff\xff",                       # absurd length prefix
+            struct.pack(">I", 3) + b"\x93\x01",        # truncated payload
+            b"GET / HTTP/1.1\r\n\r\n",                 # wrong protocol
+        ):
+            s = socklib.create_connection(("127.0.0.1", server.port), timeout=5)
+            s.sendall(garbage)
+            s.close()
+        # oversized frame header (>64 MiB declared)
+        s = socklib.create_connection(("127.0.0.1", server.port), timeout=5)
+        s.sendall(struct.pack(">I", 200 << 20))
+        s.close()
+        assert good.get("Thing", "keep")["metadata"]["name"] == "keep"
+        good.create(_obj("still-works"))
+        assert len(good.list("Thing")) == 2
+    finally:
+        good.close()
+        server.stop()

@@ -336,7 +336,32 @@ def scenario_preempt(args) -> dict:
             lat.append((time.perf_counter() - t0) * 1000)
     finally:
         c.stop()
-    return {"scenario": "preempt", "preemption_latency": _lat_stats(lat)}
+
+    # plan 2: whole-GPU preemption — an SPX pod displaces 8 low-priority
+    # CPX pods (evict all + nominate GPU + drain + mode flip + place)
+    c = _mk_cluster(num_gpus=2)
+    lat2 = []
+    try:
+        from instaslice_amd.api.types import new_pod
+
+        for rnd in range(args.pods or 5):
+            for i in range(8):
+                c.store.create(new_pod(f"v{rnd}-{i}", profile="cpx-1x36",
+                                       priority=1))
+                c.wait_pod_scheduled(f"v{rnd}-{i}", timeout=15.0)
+            t0 = time.perf_counter()
+            c.store.create(new_pod(f"big{rnd}", profile="spx-8x288",
+                                   priority=10))
+            c.wait_pod_scheduled(f"big{rnd}", timeout=30.0)
+            lat2.append((time.perf_counter() - t0) * 1000)
+            c.delete_pod(f"big{rnd}")
+            c.wait_pod_gone(f"big{rnd}")
+            c.wait_pod_unallocated(f"big{rnd}")
+    finally:
+        c.stop()
+    return {"scenario": "preempt",
+            "preemption_latency": _lat_stats(lat),
+            "whole_gpu_preemption_latency": _lat_stats(lat2)}
 
 
 SCENARIOS = {

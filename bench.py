@@ -391,11 +391,11 @@ def main() -> int:
     # initialization warm (untimed, before the contractual W warmup steps):
     # a cold process under-reports ~25% for the first seconds (allocator,
     # TCP stack, interpreter caches — measured 557 -> 824 pods/s across
-    # back-to-back runs on one box); run lifecycles until the machinery is
-    # warm, like a training bench warms its JIT before the timed window
+    # back-to-back runs on one box; ~6 s of lifecycles closes most of it);
+    # run lifecycles until warm, as a training bench warms its JIT
     t_init = time.perf_counter()
     i_init = 0
-    while time.perf_counter() - t_init < 2.0 and i_init < 2000:
+    while time.perf_counter() - t_init < 6.0 and i_init < 6000:
         step(-1 - i_init, record=False)
         i_init += 1
 

@@ -395,7 +395,7 @@ def main() -> int:
     # run lifecycles until warm, as a training bench warms its JIT
     t_init = time.perf_counter()
     i_init = 0
-    while time.perf_counter() - t_init < 6.0 and i_init < 6000:
+    while time.perf_counter() - t_init < 2.0 and i_init < 2000:
         step(-1 - i_init, record=False)
         i_init += 1
 

@@ -182,3 +182,78 @@ def test_cli_controlplane_one_command():
             agent.wait(timeout=5)
         p.terminate()
         p.wait(timeout=10)
+
+
+def test_cli_top_describe_cordon():
+    """Operator commands: top streams events, describe shows pod+events,
+    cordon/uncordon gates placement — all over the CLI."""
+    from instaslice_amd.store.native import NativeStoreServer, stored_available
+    from instaslice_amd.store.netstore import NetStoreClient
+
+    if not stored_available():
+        pytest.skip("instaslice-stored not built")
+    server = NativeStoreServer().start()
+    addr = f"127.0.0.1:{server.port}"
+    procs = []
+
+    def spawn(*cmd):
+        p = subprocess.Popen([sys.executable, "-m", "instaslice_amd", *cmd],
+                             stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                             text=True)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("controller", "--store", addr, "--grace", "0",
+              "--metrics-port", "0")
+        spawn("daemonset", "--store", addr, "--node-name", "opnode",
+              "--fake", "1", "--metrics-port", "0")
+        top = spawn("top", "--store", addr, "--interval", "1")
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "submit", "--store", addr,
+             "--name", "oppod", "--profile", "cpx-1x36", "--wait",
+             "--timeout", "30"], capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stdout + out.stderr
+
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "describe", "--store",
+             addr, "--name", "oppod"], capture_output=True, text=True,
+            timeout=30)
+        desc = json.loads(out.stdout)
+        assert desc["pod"]["gated"] is False
+        assert desc["allocation"]["profile"] == "cpx-1x36"
+        deadline = time.monotonic() + 5
+        reasons = set()
+        while time.monotonic() < deadline:
+            out = subprocess.run(
+                [sys.executable, "-m", "instaslice_amd", "describe", "--store",
+                 addr, "--name", "oppod"], capture_output=True, text=True,
+                timeout=30)
+            reasons = {e["reason"] for e in json.loads(out.stdout)["events"]}
+            if {"Placed", "PartitionReady"} <= reasons:
+                break
+            time.sleep(0.1)
+        assert {"Placed", "PartitionReady"} <= reasons, reasons
+
+        out = subprocess.run(
+            [sys.executable, "-m", "instaslice_amd", "cordon", "--store",
+             addr, "--node", "opnode"], capture_output=True, text=True,
+            timeout=30)
+        assert "cordoned" in out.stdout
+        c = NetStoreClient("127.0.0.1", server.port)
+        assert c.get("Instaslice", "opnode",
+                     "instaslice-system")["spec"]["cordoned"] is True
+        c.close()
+
+        top.terminate()
+        top_out, _ = top.communicate(timeout=10)
+        assert "pod default/oppod" in top_out
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
+        server.stop()

@@ -25,7 +25,7 @@ import threading
 import time
 from typing import Dict, List, Optional
 
-from instaslice_amd import POD_RESOURCE_PREFIX, RESOURCE_PREFIX
+from instaslice_amd import GATE_NAME, POD_RESOURCE_PREFIX, RESOURCE_PREFIX
 from instaslice_amd.api.events import emit
 from instaslice_amd.api.types import (
     AllocationStatus,
@@ -43,7 +43,8 @@ from instaslice_amd.smi.base import (
     SmiNotSupported,
     SmiPermission,
 )
-from instaslice_amd.store.memstore import AlreadyExists, MemStore, NotFound
+from instaslice_amd.store.memstore import (AlreadyExists, Conflict,
+                                            MemStore, NotFound)
 from instaslice_amd.utils import get_logger
 
 INSTASLICE_NS = "instaslice-system"
@@ -58,6 +59,7 @@ class NodeAgent:
         node_name: str,
         reset_mode_on_empty: bool = False,
         heartbeat_every_s: float = 5.0,
+        fast_ungate: bool = True,
     ) -> None:
         self.store = store
         self.smi = smi
@@ -67,6 +69,9 @@ class NodeAgent:
         # sticky when a GPU drains (the next same-profile pod lands with zero
         # reconfiguration). Set True for reference-parity teardown.
         self.reset_mode_on_empty = reset_mode_on_empty
+        # data-plane admission completion (see _commit_prepare); turn off
+        # for strict reference-parity flow (controller ungates everything)
+        self.fast_ungate = fast_ungate
         self.log = get_logger(f"agent.{node_name}")
         # cached enumeration: gpu uuid -> PhysicalGpu (refreshed only after a
         # mode change of that GPU)
@@ -283,9 +288,15 @@ class NodeAgent:
         """ONE batched store round-trip for the whole create-path commit:
         pod ConfigMap (visible-devices contract, reference createConfigMap
         instaslice_daemonset.go:796-818 -> ROCR_/HIP_ here) + node capacity
-        pin (createInstaSliceResource, :277-300) + CR status flip creating ->
-        created with the Prepared entry. Returns False if the allocation
-        changed under us (raced teardown)."""
+        pin (createInstaSliceResource, :277-300) + CR status flip + the
+        FAST-UNGATE: for non-gang pods the same batch removes the pod's
+        scheduling gate and marks the allocation `ungated` directly — the
+        data plane completes the admission it just realized, saving the
+        created-event -> controller -> ungate round (two cross-process event
+        hops). Gang members stay `created` so the controller's group barrier
+        decides when they ungate. If the pod's gate isn't exactly ours the
+        guarded patch conflicts and the classic controller path takes over.
+        Returns False if the allocation changed under us (raced teardown)."""
         part_uuid, prep = next(iter(prepared_entry.items()))
         gpu_uuid = alloc["gpuUUID"]
         g = self._gpus[gpu_uuid]
@@ -306,6 +317,10 @@ class NodeAgent:
             },
         }
         pu = alloc["podUUID"]
+        # gang members must NOT self-ungate (controller barrier owns that)
+        fast_ungate = self.fast_ungate and not alloc.get("group")
+        target_status = (AllocationStatus.UNGATED if fast_ungate
+                         else AllocationStatus.CREATED)
         res = self.store.batch([
             {"verb": "create", "obj": cm},
             {"verb": "patch", "kind": "Node", "name": self.node_name,
@@ -321,7 +336,7 @@ class NodeAgent:
                   "value": AllocationStatus.CREATING},
                  {"op": "set",
                   "path": ["spec", "allocations", pu, "allocationStatus"],
-                  "value": AllocationStatus.CREATED},
+                  "value": target_status},
                  {"op": "merge", "path": ["spec", "prepared"],
                   "value": prepared_entry},
                  {"op": "set", "path": ["spec", "gpus", gpu_uuid, "computeMode"],
@@ -332,11 +347,37 @@ class NodeAgent:
                   "path": ["spec", "gpus", gpu_uuid, "usedOrdinals"],
                   "value": alloc["ordinal"]},
              ]},
+            *([{"verb": "patch", "kind": "Pod", "name": alloc["podName"],
+                "namespace": alloc["namespace"], "ops": [
+                    {"op": "test", "path": ["spec", "schedulingGates"],
+                     "value": [{"name": GATE_NAME}]},
+                    {"op": "set", "path": ["spec", "schedulingGates"],
+                     "value": []},
+                    {"op": "set", "path": ["status", "phase"],
+                     "value": "Pending"},
+                    {"op": "set", "path": ["status", "conditions"],
+                     "value": [{"type": "PodScheduled", "status": "True",
+                                "message": "ungated"}]},
+                ]}] if fast_ungate else []),
         ], quiet=True)
         if not res[0]["ok"] and res[0]["error"]["type"] != "AlreadyExists":
             self.log.warning("configmap create failed: %s", res[0]["error"])
         if not res[1]["ok"]:
             self.log.warning("node capacity pin failed: %s", res[1]["error"])
+        if fast_ungate and not res[3]["ok"]:
+            # unusual gate set (extra gates / already ungated): hand the
+            # admission back to the controller by downgrading to `created`
+            try:
+                self.store.patch("Instaslice", self.node_name, INSTASLICE_NS, [
+                    {"op": "test",
+                     "path": ["spec", "allocations", pu, "allocationStatus"],
+                     "value": AllocationStatus.UNGATED},
+                    {"op": "set",
+                     "path": ["spec", "allocations", pu, "allocationStatus"],
+                     "value": AllocationStatus.CREATED},
+                ], quiet=True)
+            except (Conflict, NotFound):
+                pass
         commit = res[2]
         if not commit["ok"]:
             # Conflict: status moved (pod deleted mid-create) — the event for

@@ -640,6 +640,16 @@ class PodController:
             return Result()
 
         if not pod_is_gated(pod):
+            # allocation-latency bookkeeping also lands here when the AGENT
+            # fast-ungated the pod (data-plane admission completion) — the
+            # ungate event reaches us via the pod watch either way
+            t0 = self._pending_since.pop(uid, None)
+            if t0 is not None:
+                dt = time.monotonic() - t0
+                self.alloc_latency_s.append(dt)
+                from instaslice_amd.metrics import get_metrics
+
+                get_metrics().allocation_latency(dt)
             return Result()
 
         # admission path

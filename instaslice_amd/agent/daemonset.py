@@ -144,6 +144,11 @@ class NodeAgent:
                 "index": g.index,
             }
         spec["placements"] = catalog.to_dict()
+        # teardown ownership: when the agent must touch the device on drain
+        # (mode reset), the controller leaves teardown to the agent's
+        # deleted-status protocol; otherwise the controller fast-paths the
+        # whole cleanup in its own batch (reconciler._reconcile teardown)
+        spec["agentManagedTeardown"] = bool(self.reset_mode_on_empty)
         cr["status"]["processed"] = "true"
 
         try:
@@ -163,6 +168,7 @@ class NodeAgent:
             def refresh(obj: dict) -> Optional[dict]:
                 obj["spec"]["gpuUuids"] = spec["gpuUuids"]
                 obj["spec"]["placements"] = spec["placements"]
+                obj["spec"]["agentManagedTeardown"] = spec["agentManagedTeardown"]
                 for uuid, gd in spec["gpus"].items():
                     old = obj["spec"].setdefault("gpus", {}).get(uuid)
                     if old:
@@ -380,9 +386,21 @@ class NodeAgent:
                 pass
         commit = res[2]
         if not commit["ok"]:
-            # Conflict: status moved (pod deleted mid-create) — the event for
-            # that change re-reconciles us with fresh state
+            # Conflict: status moved (pod deleted mid-create). Undo our own
+            # ConfigMap + capacity pin — under the controller's fast
+            # teardown nobody else will clean them up (the allocation may
+            # already be gone from the CR)
             self.log.debug("prepare commit superseded: %s", commit["error"]["msg"])
+            self.store.batch([
+                {"verb": "delete", "kind": "ConfigMap",
+                 "name": alloc["podName"], "namespace": alloc["namespace"]},
+                {"verb": "patch", "kind": "Node", "name": self.node_name,
+                 "namespace": "", "ops": [
+                     {"op": "delete",
+                      "path": ["status", "capacity",
+                               POD_RESOURCE_PREFIX + alloc["podName"]]},
+                 ]},
+            ], quiet=True)
             return False
         emit(self.store,
              {"kind": "Pod", "namespace": alloc["namespace"],

@@ -611,7 +611,48 @@ class PodController:
             found = self._find_allocation(uid, self._crs())
             if found:
                 cr, alloc = found
-                if alloc["allocationStatus"] != AllocationStatus.DELETED:
+                if not (cr["spec"].get("agentManagedTeardown")):
+                    # FAST TEARDOWN: the agent's drain work is pure store
+                    # bookkeeping unless it must reset the GPU mode (it
+                    # advertises that via spec.agentManagedTeardown), so do
+                    # the whole cleanup here in ONE batch — ConfigMap,
+                    # capacity pin, allocation/prepared/ordinal — instead of
+                    # the two-phase deleted-status handoff (saves one
+                    # cross-process event hop + one round-trip per drain)
+                    node = cr["metadata"]["name"]
+                    part_uuids = [
+                        k for k, v in (cr["spec"].get("prepared") or {}).items()
+                        if v["podUUID"] == uid
+                    ]
+                    reqs.append(
+                        {"verb": "delete", "kind": "ConfigMap",
+                         "name": alloc["podName"],
+                         "namespace": alloc["namespace"]})
+                    reqs.append(
+                        {"verb": "patch", "kind": "Node", "name": node,
+                         "namespace": "", "ops": [
+                             {"op": "delete",
+                              "path": ["status", "capacity",
+                                       "org.instaslice/" + alloc["podName"]]},
+                         ]})
+                    reqs.append(
+                        {"verb": "patch", "kind": "Instaslice", "name": node,
+                         "namespace": INSTASLICE_NS, "ops": [
+                             {"op": "test",
+                              "path": ["spec", "allocations", uid, "podUUID"],
+                              "value": uid},
+                             {"op": "delete",
+                              "path": ["spec", "allocations", uid]},
+                             {"op": "remove_from_set",
+                              "path": ["spec", "gpus", alloc["gpuUUID"],
+                                       "usedOrdinals"],
+                              "value": alloc["ordinal"]},
+                         ] + [
+                             {"op": "delete",
+                              "path": ["spec", "prepared", puid]}
+                             for puid in part_uuids
+                         ]})
+                elif alloc["allocationStatus"] != AllocationStatus.DELETED:
                     reqs.append(
                         {"verb": "patch", "kind": "Instaslice",
                          "name": cr["metadata"]["name"],
@@ -624,6 +665,13 @@ class PodController:
                                        "allocationStatus"],
                               "value": AllocationStatus.DELETED},
                          ]})
+            fast_cleanup_idx = next(
+                (i for i, r in enumerate(reqs)
+                 if r.get("kind") == "Instaslice"
+                 and any(op["op"] == "delete" and op["path"][:2] ==
+                         ["spec", "allocations"] for op in r["ops"])),
+                None,
+            )
             if FINALIZER_NAME in (md.get("finalizers") or []):
                 # guard on our view: an unconditional strip would bump the
                 # rv (and emit MODIFIED) even when already stripped —
@@ -636,7 +684,14 @@ class PodController:
                           "value": FINALIZER_NAME},
                      ]})
             if reqs:
-                self.store.batch(reqs, quiet=True)
+                res = self.store.batch(reqs, quiet=True)
+                if fast_cleanup_idx is not None and res[fast_cleanup_idx]["ok"]:
+                    from instaslice_amd.metrics import get_metrics
+
+                    get_metrics().allocation("deleted")
+                    emit(self.store, pod, "PartitionReleased",
+                         f"partition ordinal {alloc['ordinal']} on "
+                         f"{alloc['gpuUUID'][:8]} released")
             return Result()
 
         if not pod_is_gated(pod):

@@ -50,10 +50,33 @@ class _EventSink:
 
         self._q: "queue.Queue" = queue.Queue(maxsize=4096)
         self._lru: "collections.OrderedDict" = collections.OrderedDict()
+        # TCP stores get a DEDICATED connection per client: the store serves
+        # each connection's requests serially, so event writes on the
+        # caller's connection would queue AHEAD of its latency-critical
+        # reconcile calls (measured as controller workers blocked in _call
+        # behind the sink)
+        self._conns: dict = {}
         self.dropped = 0
         self._thread = threading.Thread(target=self._run, daemon=True,
                                         name="event-sink")
         self._thread.start()
+
+    def _writer_store(self, store):
+        host = getattr(store, "host", None)
+        port = getattr(store, "port", None)
+        if host is None or port is None:
+            return store  # in-process store: use directly
+        key = (host, port)
+        conn = self._conns.get(key)
+        if conn is None or getattr(conn, "_closed", False):
+            from instaslice_amd.store.netstore import NetStoreClient
+
+            try:
+                conn = NetStoreClient(host, port, reconnect=True)
+            except OSError:
+                return store
+            self._conns[key] = conn
+        return conn
 
     # under sustained churn Normal events are sampled away once the queue
     # backs up (k8s recorders rate-limit the same way); Warnings always keep
@@ -73,8 +96,8 @@ class _EventSink:
         while True:
             store, ref, ns, name, reason, message, type_, component = self._q.get()
             try:
-                _write_event(store, ref, ns, name, reason, message,
-                             type_, component)
+                _write_event(self._writer_store(store), ref, ns, name,
+                             reason, message, type_, component)
                 self._lru[(ns, name)] = store
                 self._lru.move_to_end((ns, name))
                 while len(self._lru) > self.MAX_LIVE_EVENTS:

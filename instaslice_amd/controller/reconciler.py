@@ -286,6 +286,12 @@ class PodController:
             self._unschedulable_keys.add(key)
         else:
             self._unschedulable_keys.discard(key)
+            # common case: the pod was never marked — skip the
+            # read-modify-write entirely (one store GET per placement
+            # otherwise; stale-view miss just leaves a cosmetic annotation
+            # that the next event-driven pass clears)
+            if UNSCHEDULABLE_ANNOTATION not in (md.get("annotations") or {}):
+                return
 
         def mut(p: dict) -> Optional[dict]:
             ann = p["metadata"].setdefault("annotations", {})

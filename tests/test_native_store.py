@@ -4,6 +4,7 @@ against BOTH backends via the parametrized fixture; the native daemon is the
 control plane's scale-out path (store/csrc/stored_main.cpp), MemStore the
 reference semantics (store/memstore.py)."""
 
+import os
 import threading
 import time
 
@@ -463,3 +464,68 @@ def test_daemon_survives_protocol_garbage():
     finally:
         good.close()
         server.stop()
+
+
+from hypothesis import HealthCheck, given, settings
+from hypothesis import strategies as st
+
+_keys = st.sampled_from(["a", "b", "c", "list", "allocations", "x"])
+_vals = st.one_of(st.integers(-5, 5), st.text(max_size=4), st.booleans(),
+                  st.none(), st.dictionaries(_keys, st.integers(0, 3),
+                                             max_size=2))
+_paths = st.lists(_keys, min_size=1, max_size=3)
+_ops = st.lists(st.one_of(
+    st.fixed_dictionaries({"op": st.just("set"), "path": _paths,
+                           "value": _vals}),
+    st.fixed_dictionaries({"op": st.just("merge"), "path": _paths,
+                           "value": st.dictionaries(_keys, _vals, max_size=2)}),
+    st.fixed_dictionaries({"op": st.just("delete"), "path": _paths}),
+    st.fixed_dictionaries({"op": st.just("add_to_set"), "path": _paths,
+                           "value": st.integers(0, 5)}),
+    st.fixed_dictionaries({"op": st.just("remove_from_set"), "path": _paths,
+                           "value": st.integers(0, 5)}),
+    st.fixed_dictionaries({"op": st.just("test"), "path": _paths,
+                           "value": _vals}),
+    st.fixed_dictionaries({"op": st.just("test"), "path": _paths,
+                           "absent": st.just(True)}),
+), min_size=1, max_size=6)
+
+
+@needs_stored
+@settings(max_examples=int(os.environ.get(
+              "INSTASLICE_FUZZ_EXAMPLES", "15")),
+          deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(batches=st.lists(_ops, min_size=1, max_size=5))
+def test_patch_grammar_differential_fuzz(batches):
+    """Differential fuzzing: the SAME random patch sequences applied to the
+    Python MemStore and the C++ daemon must produce byte-identical objects
+    and identical error types — the strongest form of the parity contract."""
+    py_server = StoreServer().start()
+    nat_server = NativeStoreServer().start()
+    py = NetStoreClient("127.0.0.1", py_server.port)
+    nat = NetStoreClient("127.0.0.1", nat_server.port)
+    try:
+        seed = {"apiVersion": "v1", "kind": "Thing",
+                "metadata": {"name": "f", "namespace": ""},
+                "spec": {"allocations": {}, "list": [1, 2]}}
+        py.create(seed)
+        nat.create(seed)
+        for ops in batches:
+            ery = ern = None
+            try:
+                py.patch("Thing", "f", "", ops)
+            except Exception as e:  # noqa: BLE001
+                ery = type(e).__name__
+            try:
+                nat.patch("Thing", "f", "", ops)
+            except Exception as e:  # noqa: BLE001
+                ern = type(e).__name__
+            assert ery == ern, f"error divergence {ery} vs {ern} on {ops}"
+        a, b = py.get("Thing", "f"), nat.get("Thing", "f")
+        assert a == b, f"state divergence:\n{a}\n{b}"
+    finally:
+        py.close()
+        nat.close()
+        py_server.stop()
+        nat_server.stop()

@@ -454,13 +454,17 @@ class Store {
     }
 
     if (path.empty()) throw StoreError{"Error", "empty patch path"};
-    // navigate to parent, COW-cloning each spine level
+    // navigate to parent, COW-cloning each spine level; lists are leaf
+    // containers only — traversing into one errors (MemStore parity,
+    // pinned by the differential fuzz test)
     Value* node = &obj;
     cow_map(*node);
     for (size_t k = 0; k + 1 < path.size(); ++k) {
       if (!path[k].is_str()) throw StoreError{"Error", "non-string path"};
       Value* nxt = node->find(path[k].s);
-      if (!nxt || (!nxt->is_map() && !nxt->is_arr()))
+      if (nxt && nxt->is_arr())
+        throw StoreError{"Error", "patch path traverses a list"};
+      if (!nxt || !nxt->is_map())
         nxt = &node->setkey(path[k].s, Value::map());
       cow_map(*nxt);
       node = nxt;

@@ -85,11 +85,15 @@ def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
             elif missing or node != op.get("value"):
                 raise Conflict(f"patch test: {path} != {op.get('value')!r}")
             continue
-        # navigate to parent, creating dicts along the way
+        # navigate to parent, creating dicts along the way; lists are leaf
+        # containers only (add_to_set/remove_from_set) — traversing INTO one
+        # is a malformed path and errors identically on every backend
         node = obj
         for p in path[:-1]:
             nxt = node.get(p) if isinstance(node, dict) else None
-            if not isinstance(nxt, (dict, list)):
+            if isinstance(nxt, list):
+                raise ValueError(f"patch path traverses a list at {p!r}")
+            if not isinstance(nxt, dict):
                 nxt = {}
                 node[p] = nxt
             node = nxt

@@ -491,41 +491,53 @@ _ops = st.lists(st.one_of(
 ), min_size=1, max_size=6)
 
 
+_fuzz_env = {}
+_fuzz_n = [0]
+
+
+def _fuzz_backends():
+    """Shared servers for the differential fuzz (per-example server spawn
+    exhausted fds/threads at high example counts)."""
+    if not _fuzz_env:
+        py_server = StoreServer().start()
+        nat_server = NativeStoreServer().start()
+        _fuzz_env["servers"] = (py_server, nat_server)
+        _fuzz_env["py"] = NetStoreClient("127.0.0.1", py_server.port)
+        _fuzz_env["nat"] = NetStoreClient("127.0.0.1", nat_server.port)
+    return _fuzz_env["py"], _fuzz_env["nat"]
+
+
 @needs_stored
 @settings(max_examples=int(os.environ.get(
-              "INSTASLICE_FUZZ_EXAMPLES", "15")),
+              "INSTASLICE_FUZZ_EXAMPLES", "25")),
           deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(batches=st.lists(_ops, min_size=1, max_size=5))
 def test_patch_grammar_differential_fuzz(batches):
     """Differential fuzzing: the SAME random patch sequences applied to the
     Python MemStore and the C++ daemon must produce byte-identical objects
-    and identical error types — the strongest form of the parity contract."""
-    py_server = StoreServer().start()
-    nat_server = NativeStoreServer().start()
-    py = NetStoreClient("127.0.0.1", py_server.port)
-    nat = NetStoreClient("127.0.0.1", nat_server.port)
-    try:
-        seed = {"apiVersion": "v1", "kind": "Thing",
-                "metadata": {"name": "f", "namespace": ""},
-                "spec": {"allocations": {}, "list": [1, 2]}}
-        py.create(seed)
-        nat.create(seed)
-        for ops in batches:
-            ery = ern = None
-            try:
-                py.patch("Thing", "f", "", ops)
-            except Exception as e:  # noqa: BLE001
-                ery = type(e).__name__
-            try:
-                nat.patch("Thing", "f", "", ops)
-            except Exception as e:  # noqa: BLE001
-                ern = type(e).__name__
-            assert ery == ern, f"error divergence {ery} vs {ern} on {ops}"
-        a, b = py.get("Thing", "f"), nat.get("Thing", "f")
-        assert a == b, f"state divergence:\n{a}\n{b}"
-    finally:
-        py.close()
-        nat.close()
-        py_server.stop()
-        nat_server.stop()
+    and identical error types — the strongest form of the parity contract.
+    (Found for real: list-traversing paths used to error on Python and
+    silently clobber on C++.)"""
+    py, nat = _fuzz_backends()
+    _fuzz_n[0] += 1
+    name = f"f{_fuzz_n[0]}"
+    seed = {"apiVersion": "v1", "kind": "Thing",
+            "metadata": {"name": name, "namespace": ""},
+            "spec": {"allocations": {}, "list": [1, 2]}}
+    py.create(seed)
+    nat.create(seed)
+    for ops in batches:
+        ery = ern = None
+        try:
+            py.patch("Thing", name, "", ops)
+        except Exception as e:  # noqa: BLE001
+            ery = type(e).__name__
+        try:
+            nat.patch("Thing", name, "", ops)
+        except Exception as e:  # noqa: BLE001
+            ern = type(e).__name__
+        assert ery == ern, f"error divergence {ery} vs {ern} on {ops}"
+    a, b = py.get("Thing", name), nat.get("Thing", name)
+    b = dict(b)
+    assert a == b, f"state divergence:\n{a}\n{b}"

@@ -15,6 +15,9 @@
                                         [--node NODE] [--wait]
     python -m instaslice_amd delete     --store HOST:PORT --name P
     python -m instaslice_amd status     --store HOST:PORT
+    python -m instaslice_amd describe   --store HOST:PORT --name P
+    python -m instaslice_amd top        --store HOST:PORT [--interval S]
+    python -m instaslice_amd cordon     --store HOST:PORT --node N [--uncordon]
     python -m instaslice_amd payload    [info|vecadd|membw|busy|census] ...
 
 NODE_NAME env is honored for the daemonset (downward-API parity with

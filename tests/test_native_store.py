@@ -541,3 +541,90 @@ def test_patch_grammar_differential_fuzz(batches):
     a, b = py.get("Thing", name), nat.get("Thing", name)
     b = dict(b)
     assert a == b, f"state divergence:\n{a}\n{b}"
+
+
+_verb_ops = st.lists(st.one_of(
+    st.fixed_dictionaries({"verb": st.just("create"),
+                           "fin": st.booleans(),
+                           "spec": st.dictionaries(_keys, _vals, max_size=2)}),
+    st.fixed_dictionaries({"verb": st.just("update"),
+                           "stale": st.booleans(),
+                           "spec": st.dictionaries(_keys, _vals, max_size=2)}),
+    st.fixed_dictionaries({"verb": st.just("delete")}),
+    st.fixed_dictionaries({"verb": st.just("strip_finalizers")}),
+    st.fixed_dictionaries({"verb": st.just("patch"), "ops": _ops}),
+), min_size=1, max_size=10)
+
+
+@needs_stored
+@settings(max_examples=int(os.environ.get(
+              "INSTASLICE_FUZZ_EXAMPLES", "25")),
+          deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(seq=_verb_ops)
+def test_verb_lifecycle_differential_fuzz(seq):
+    """Differential fuzzing over the FULL verb set, including finalizer
+    two-phase deletion and stale-resourceVersion conflicts: both backends
+    must expose identical state and identical error types after every
+    step. (rv sequences are deterministic, so stale-rv updates conflict —
+    or not — identically.)"""
+    py, nat = _fuzz_backends()
+    _fuzz_n[0] += 1
+    name = f"v{_fuzz_n[0]}"
+
+    def run(client, op):
+        if op["verb"] == "create":
+            obj = {"apiVersion": "v1", "kind": "VThing",
+                   "metadata": {"name": name, "namespace": ""},
+                   "spec": dict(op["spec"])}
+            if op["fin"]:
+                obj["metadata"]["finalizers"] = ["org.instaslice/accelarator"]
+            return client.create(obj)
+        if op["verb"] == "update":
+            cur = client.get("VThing", name, "")
+            cur["spec"] = dict(op["spec"])
+            if op["stale"]:
+                cur["metadata"]["resourceVersion"] = "0"
+            return client.update(cur)
+        if op["verb"] == "delete":
+            return client.delete("VThing", name, "")
+        if op["verb"] == "strip_finalizers":
+            cur = client.get("VThing", name, "")
+            cur["metadata"]["finalizers"] = []
+            return client.update(cur)
+        if op["verb"] == "patch":
+            return client.patch("VThing", name, "", op["ops"])
+        raise AssertionError(op)
+
+    for op in seq:
+        ery = ern = None
+        try:
+            run(py, op)
+        except Exception as e:  # noqa: BLE001
+            ery = type(e).__name__
+        try:
+            run(nat, op)
+        except Exception as e:  # noqa: BLE001
+            ern = type(e).__name__
+        assert ery == ern, f"error divergence {ery} vs {ern} on {op}"
+        sa = sb = None
+        try:
+            sa = py.get("VThing", name, "")
+        except NotFound:
+            pass
+        try:
+            sb = nat.get("VThing", name, "")
+        except NotFound:
+            pass
+        # deletionTimestamp is wall-clock and may differ by nanoseconds;
+        # compare it by presence, everything else exactly
+        def norm(o):
+            if o is None:
+                return None
+            o = dict(o)
+            md = dict(o["metadata"])
+            if md.get("deletionTimestamp"):
+                md["deletionTimestamp"] = "SET"
+            o["metadata"] = md
+            return o
+        assert norm(sa) == norm(sb), f"state divergence after {op}:\n{sa}\n{sb}"

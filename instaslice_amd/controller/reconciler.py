@@ -27,6 +27,7 @@ import time
 import zlib
 from typing import List, Optional, Tuple
 
+from instaslice_amd import FINALIZER_NAME, POD_RESOURCE_PREFIX
 from instaslice_amd.api.events import emit
 from instaslice_amd.api.types import (
     AllocationDetails,
@@ -608,11 +609,9 @@ class PodController:
                 elapsed = time.time() - float(md["deletionTimestamp"])
                 if elapsed < self.teardown_grace_s:
                     return Result(requeue_after=self.teardown_grace_s - elapsed)
-            # ONE batched round-trip: mark the allocation deleted (agent
-            # tears it down on that event) AND strip our finalizer — the
-            # store drops the pod once no finalizers remain
-            from instaslice_amd import FINALIZER_NAME
-
+            # ONE batched round-trip: full cleanup (or mark-deleted for
+            # agent-managed teardown) AND strip our finalizer — the store
+            # drops the pod once no finalizers remain
             reqs = []
             found = self._find_allocation(uid, self._crs())
             if found:
@@ -639,7 +638,7 @@ class PodController:
                          "namespace": "", "ops": [
                              {"op": "delete",
                               "path": ["status", "capacity",
-                                       "org.instaslice/" + alloc["podName"]]},
+                                       POD_RESOURCE_PREFIX + alloc["podName"]]},
                          ]})
                     reqs.append(
                         {"verb": "patch", "kind": "Instaslice", "name": node,

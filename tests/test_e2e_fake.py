@@ -588,3 +588,40 @@ def test_whole_gpu_preemption_for_bigger_profile():
         assert not (cr["spec"].get("nominations") or {})
     finally:
         c.stop()
+
+
+def test_agent_managed_teardown_resets_mode():
+    """reset_mode_on_empty=True: the agent advertises agentManagedTeardown,
+    the controller falls back to the two-phase deleted-status protocol, the
+    agent tears down AND returns the drained GPU to SPX/NPS1 (reference
+    ci/gi Destroy analog, instaslice_daemonset.go:377-413)."""
+    c = Cluster(teardown_grace_s=0.0, reset_mode_on_empty=True)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        cr = c.store.get("Instaslice", "node-0", "instaslice-system")
+        assert cr["spec"]["agentManagedTeardown"] is True
+        c.submit_pod("tp", "cpx-1x36")  # forces SPX -> CPX flip
+        c.wait_pod_scheduled("tp")
+        gpu = next(iter(c.store.get("Instaslice", "node-0",
+                                    "instaslice-system")["spec"]["gpus"]))
+        assert c.agents["node-0"].smi.get_compute_partition(gpu) == "CPX"
+        c.delete_pod("tp")
+        c.wait_pod_gone("tp")
+        c.wait_allocations_empty("node-0")
+        # drained GPU returns to SPX (reference-parity teardown)
+        import time as _t
+
+        deadline = _t.monotonic() + 10
+        while _t.monotonic() < deadline:
+            if c.agents["node-0"].smi.get_compute_partition(gpu) == "SPX":
+                break
+            _t.sleep(0.02)
+        assert c.agents["node-0"].smi.get_compute_partition(gpu) == "SPX"
+        # ConfigMap cleaned by the agent's teardown branch
+        from instaslice_amd.store.memstore import NotFound
+
+        with pytest.raises(NotFound):
+            c.store.get("ConfigMap", "tp", "default")
+    finally:
+        c.stop()

@@ -625,3 +625,86 @@ def test_agent_managed_teardown_resets_mode():
             c.store.get("ConfigMap", "tp", "default")
     finally:
         c.stop()
+
+
+def test_fast_ungate_falls_back_on_foreign_gates():
+    """A pod carrying an EXTRA scheduling gate: the agent's guarded
+    fast-ungate patch conflicts, downgrades to `created`, and the
+    controller's fallback removes only OUR gate (the foreign gate stays —
+    its owner decides)."""
+    from instaslice_amd.api.types import new_pod
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        p = new_pod("fg", profile="cpx-1x36")
+        p["spec"]["schedulingGates"].append({"name": "someone.else/gate"})
+        c.store.create(p)
+        import time as _t
+
+        deadline = _t.monotonic() + 10
+        alloc = None
+        while _t.monotonic() < deadline:
+            allocs = c.allocations("node-0")
+            alloc = next((a for a in allocs.values()
+                          if a["podName"] == "fg"), None)
+            if alloc and alloc["allocationStatus"] == "ungated":
+                break
+            _t.sleep(0.02)
+        assert alloc and alloc["allocationStatus"] == "ungated", alloc
+        pod = c.store.get("Pod", "fg", "default")
+        assert pod["spec"]["schedulingGates"] == [
+            {"name": "someone.else/gate"}], pod["spec"]["schedulingGates"]
+    finally:
+        c.stop()
+
+
+def test_mid_prepare_deletion_leaves_no_orphans():
+    """Delete a pod WHILE the agent is inside the (slow) mode set: the
+    commit patch conflicts against the fast-teardown cleanup and the agent
+    undoes its own ConfigMap + capacity pin — nothing orphaned."""
+    from instaslice_amd.store.memstore import NotFound
+
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0",
+                                    compute_set_latency_s=0.4))
+    c.start()
+    try:
+        c.submit_pod("mid", "cpx-1x36")  # needs SPX->CPX flip (0.4s)
+        import time as _t
+
+        # wait until placed (allocation exists), then delete during the flip
+        deadline = _t.monotonic() + 10
+        while _t.monotonic() < deadline:
+            if any(a["podName"] == "mid"
+                   for a in c.allocations("node-0").values()):
+                break
+            _t.sleep(0.01)
+        c.delete_pod("mid")
+        c.wait_pod_gone("mid", timeout=10.0)
+        # after the flip completes and the dust settles: no allocation, no
+        # ConfigMap, no capacity pin
+        deadline = _t.monotonic() + 10
+        while _t.monotonic() < deadline:
+            allocs = c.allocations("node-0")
+            cm_gone = False
+            try:
+                c.store.get("ConfigMap", "mid", "default")
+            except NotFound:
+                cm_gone = True
+            node = c.store.get("Node", "node-0", "")
+            cap = node["status"].get("capacity") or {}
+            if (not any(a["podName"] == "mid" for a in allocs.values())
+                    and cm_gone
+                    and "org.instaslice/mid" not in cap):
+                break
+            _t.sleep(0.05)
+        assert not any(a["podName"] == "mid"
+                       for a in c.allocations("node-0").values())
+        with pytest.raises(NotFound):
+            c.store.get("ConfigMap", "mid", "default")
+        cap = c.store.get("Node", "node-0", "")["status"].get("capacity") or {}
+        assert "org.instaslice/mid" not in cap, cap
+    finally:
+        c.stop()

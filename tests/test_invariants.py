@@ -68,6 +68,14 @@ def _settle(c: Cluster, submitted, deleted, timeout=45.0):
 
 
 def _check_invariants(c: Cluster):
+    # 0. a pod is allocated on AT MOST one node (cross-CR uniqueness)
+    owners = {}
+    for cr in c.store.list("Instaslice"):
+        for u in (cr["spec"].get("allocations") or {}):
+            assert u not in owners, (
+                f"pod {u} allocated on both {owners[u]} and "
+                f"{cr['metadata']['name']}")
+            owners[u] = cr["metadata"]["name"]
     for cr in c.store.list("Instaslice"):
         spec = cr["spec"]
         allocs = spec.get("allocations") or {}
@@ -108,15 +116,17 @@ def _check_invariants(c: Cluster):
         assert realized_names <= caps
 
 
+@pytest.mark.parametrize("nodes", [1, 2])
 @pytest.mark.parametrize("policy", ["packed-fit", "first-fit", "spread-fit"])
 @settings(max_examples=int(os.environ.get("INSTASLICE_INVARIANT_EXAMPLES",
                                            "10")),
           deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(ops=ops_strategy)
-def test_random_lifecycle_invariants(policy, ops):
+def test_random_lifecycle_invariants(policy, nodes, ops):
     c = Cluster(teardown_grace_s=0.0, policy=policy)
-    c.add_node("node-0", FakeAmdSmi(num_gpus=2, node_name="node-0"))
+    for n in range(nodes):
+        c.add_node(f"node-{n}", FakeAmdSmi(num_gpus=2, node_name=f"node-{n}"))
     c.start()
     submitted, deleted = [], set()
     try:

@@ -598,7 +598,12 @@ class PodController:
                 self._unschedulable_keys.discard((namespace, name))
                 return Result()
         md = pod["metadata"]
-        uid = md["uid"]
+        uid = md.get("uid")
+        if not uid:
+            # malformed object (externally crafted): ignoring beats an
+            # error-backoff retry loop — nothing we can key an allocation on
+            self.log.warning("pod %s/%s has no uid; ignoring", namespace, name)
+            return Result()
 
         # teardown path (reference: instaslice_controller.go:99-142)
         if md.get("deletionTimestamp"):

@@ -708,3 +708,30 @@ def test_mid_prepare_deletion_leaves_no_orphans():
         assert "org.instaslice/mid" not in cap, cap
     finally:
         c.stop()
+
+
+def test_malformed_pod_is_ignored_not_retried():
+    """An externally crafted pod without a uid (or containers) must be
+    ignored — not spin the reconciler through error backoff forever."""
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        bad = {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": "mal", "namespace": "default",
+                         "finalizers": []},
+            "spec": {"schedulingGates": [
+                {"name": "org.instaslice/accelarator"}]},
+            "status": {"phase": "Pending", "conditions": [
+                {"type": "PodScheduled", "status": "False",
+                 "message": "blocked"}]},
+        }
+        c.store.create(bad)
+        # a good pod afterwards still flows normally (no wedged worker)
+        c.submit_pod("good", "cpx-1x36")
+        c.wait_pod_scheduled("good")
+        assert c.controller.engine.error_count == 0, (
+            "malformed pod drove reconcile errors")
+    finally:
+        c.stop()

@@ -455,6 +455,16 @@ class NodeAgent:
         requeue: Optional[float] = None
 
         for pod_uuid, alloc in sorted(allocations.items()):
+            # one externally crafted/corrupt entry must not wedge the whole
+            # node's reconcile loop (KeyError would abort every iteration
+            # through engine error backoff, forever)
+            if not isinstance(alloc, dict) or not all(
+                k in alloc for k in
+                ("allocationStatus", "gpuUUID", "ordinal", "podName",
+                 "namespace", "podUUID")
+            ):
+                self.log.warning("skipping malformed allocation %r", pod_uuid)
+                continue
             status = alloc["allocationStatus"]
             if status == AllocationStatus.CREATING:
                 needs_flip = self._needs_mode_change(alloc)

@@ -735,3 +735,21 @@ def test_malformed_pod_is_ignored_not_retried():
             "malformed pod drove reconcile errors")
     finally:
         c.stop()
+
+
+def test_malformed_allocation_entry_does_not_wedge_agent():
+    """A corrupt allocations entry (external write) is skipped; the agent
+    keeps realizing the healthy ones."""
+    c = Cluster(teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=1, node_name="node-0"))
+    c.start()
+    try:
+        c.store.patch("Instaslice", "node-0", "instaslice-system", [
+            {"op": "set", "path": ["spec", "allocations", "garbage"],
+             "value": {"allocationStatus": "creating"}},  # missing fields
+        ])
+        c.submit_pod("okpod", "cpx-1x36")
+        c.wait_pod_scheduled("okpod")
+        assert c.agents["node-0"].engine.error_count == 0
+    finally:
+        c.stop()

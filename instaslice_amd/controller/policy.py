@@ -86,6 +86,11 @@ def build_gpu_views(node_name: str, spec: dict,
     occupied: Dict[str, Set[int]] = {}
     target_modes: Dict[str, Tuple[str, str]] = {}
     for alloc in (spec.get("allocations") or {}).values():
+        # tolerate externally crafted/corrupt entries: a missing key here
+        # would wedge every consumer (controller placement, agent capacity,
+        # heartbeats) through their retry loops
+        if not isinstance(alloc, dict) or "gpuUUID" not in alloc                 or "ordinal" not in alloc:
+            continue
         occupied.setdefault(alloc["gpuUUID"], set()).add(alloc["ordinal"])
         if alloc.get("computeMode"):
             target_modes[alloc["gpuUUID"]] = (
@@ -93,6 +98,8 @@ def build_gpu_views(node_name: str, spec: dict,
                 alloc.get("memoryMode") or "NPS1",
             )
     for prep in (spec.get("prepared") or {}).values():
+        if not isinstance(prep, dict) or "parentGpuUUID" not in prep                 or "ordinal" not in prep:
+            continue
         occupied.setdefault(prep["parentGpuUUID"], set()).add(prep["ordinal"])
     # preemption nominations: a slot freed by an eviction is reserved for
     # the preemptor (k8s nominatedNodeName analog) — everyone else sees it

@@ -4,7 +4,7 @@ import time
 
 from instaslice_amd.api import events as ev_mod
 from instaslice_amd.api.events import _EventSink, emit
-from instaslice_amd.store.memstore import MemStore, NotFound
+from instaslice_amd.store.memstore import MemStore
 
 
 def _drain(sink, timeout=5.0):
@@ -47,9 +47,10 @@ def test_lru_retention_deletes_oldest():
 
 def test_normal_events_shed_under_backlog():
     sink = _fresh_sink()
-    sink.NORMAL_SHED_DEPTH = 0  # any queued item means backlog
+    # deterministic backlog: depth -1 makes qsize() > depth always true, so
+    # EVERY Normal sheds while Warnings always keep
+    sink.NORMAL_SHED_DEPTH = -1
     store = MemStore()
-    # first item occupies the queue; subsequent Normals shed, Warning keeps
     for i in range(10):
         emit(store, {"kind": "Pod", "namespace": "default", "name": f"n{i}"},
              "Placed", "x")
@@ -57,10 +58,5 @@ def test_normal_events_shed_under_backlog():
          "Boom", "x", type_="Warning")
     _drain(sink)
     names = {e["metadata"]["name"] for e in store.list("Event")}
-    assert "warn.Boom" in names
-    assert sink.dropped > 0
-    try:
-        total = len(names)
-    except NotFound:  # pragma: no cover
-        total = 0
-    assert total < 11  # most Normals were shed
+    assert names == {"warn.Boom"}, names
+    assert sink.dropped == 10

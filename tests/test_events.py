@@ -2,6 +2,8 @@
 
 import time
 
+import pytest
+
 from instaslice_amd.api import events as ev_mod
 from instaslice_amd.api.events import _EventSink, emit
 from instaslice_amd.store.memstore import MemStore
@@ -17,6 +19,14 @@ def _drain(sink, timeout=5.0):
 def _fresh_sink():
     ev_mod._sink = _EventSink()
     return ev_mod._sink
+
+
+@pytest.fixture(autouse=True)
+def _reset_sink_after():
+    # these tests tune the GLOBAL sink (shed depth, LRU cap); leave a clean
+    # default for whatever runs next
+    yield
+    ev_mod._sink = None
 
 
 def test_emit_dedup_bumps_count():

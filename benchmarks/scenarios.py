@@ -11,7 +11,8 @@
 
 All scenarios run against FakeAmdSmi by default so they execute anywhere
 (bench.py is the real-GPU entry point; scenarios quantify policy behavior).
-Run: python -m benchmarks.scenarios [--scenario all] [--pods N] [--json]
+Run: python -m benchmarks.scenarios [--scenario all] [--pods N]
+     [--policy P] [--reconfig-latency S]   (output is JSON lines)
 """
 
 from __future__ import annotations

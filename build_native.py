@@ -34,11 +34,32 @@ def pybind_includes() -> list:
     return [f"-I{pybind11.get_include()}", f"-I{sysconfig.get_paths()['include']}"]
 
 
-def newer(target: Path, sources: list) -> bool:
+def _srchash(sources: list, cmd_sig: str) -> str:
+    """Content hash of the sources + compile signature. mtime freshness was
+    dropped (advisor r1): a stale artifact with a newer mtime could silently
+    masquerade as a fresh build; a content hash is deterministic."""
+    import hashlib
+
+    h = hashlib.sha256()
+    h.update(cmd_sig.encode())
+    for s in sources:
+        h.update(Path(s).read_bytes())
+    return h.hexdigest()
+
+
+def newer(target: Path, sources: list, cmd_sig: str = "") -> bool:
+    """True iff `target` was built from exactly these source bytes."""
     if not target.exists():
         return False
-    t = target.stat().st_mtime
-    return all(t >= Path(s).stat().st_mtime for s in sources)
+    sidecar = target.with_suffix(target.suffix + ".srchash")
+    if not sidecar.exists():
+        return False
+    return sidecar.read_text().strip() == _srchash(sources, cmd_sig)
+
+
+def record_hash(target: Path, sources: list, cmd_sig: str = "") -> None:
+    sidecar = target.with_suffix(target.suffix + ".srchash")
+    sidecar.write_text(_srchash(sources, cmd_sig) + "\n")
 
 
 def run(cmd: list) -> None:
@@ -60,28 +81,32 @@ def build(force: bool = False) -> None:
     src = smi_dir / "csrc" / "partitiond_pybind.cpp"
     hdr = smi_dir / "csrc" / "partitiond_core.hpp"
     out = smi_dir / f"_partitiond{EXT}"
-    if force or not newer(out, [src, hdr]):
+    if force or not newer(out, [src, hdr], "pybind"):
         run([CLANGXX, "-shared", *common, *pybind_includes(), src, "-lamd_smi", "-o", out])
+        record_hash(out, [src, hdr], "pybind")
 
     # 2. partitiond standalone daemon
     src = smi_dir / "csrc" / "partitiond_main.cpp"
     out = bin_dir / "partitiond"
-    if force or not newer(out, [src, hdr]):
+    if force or not newer(out, [src, hdr], "daemon"):
         run([CLANGXX, *common, src, "-lamd_smi", "-o", out])
+        record_hash(out, [src, hdr], "daemon")
 
     # 3. _payload pybind module (gfx950 device code)
     src = ops_dir / "csrc" / "payload_pybind.hip"
     hdr = ops_dir / "csrc" / "payload_core.hpp"
     out = ops_dir / f"_payload{EXT}"
-    if force or not newer(out, [src, hdr]):
+    if force or not newer(out, [src, hdr], ARCH):
         run([HIPCC, f"--offload-arch={ARCH}", "-shared", *common,
              *pybind_includes(), src, "-o", out])
+        record_hash(out, [src, hdr], ARCH)
 
     # 4. instaslice-payload standalone workload binary
     src = ops_dir / "csrc" / "payload_main.hip"
     out = bin_dir / "instaslice-payload"
-    if force or not newer(out, [src, hdr]):
+    if force or not newer(out, [src, hdr], ARCH):
         run([HIPCC, f"--offload-arch={ARCH}", *common, src, "-o", out])
+        record_hash(out, [src, hdr], ARCH)
 
     # 5. instaslice-stored: native store daemon (plain host C++, g++ —
     # no ROCm dependency; runs on any node incl. the CPU test tier)
@@ -89,8 +114,9 @@ def build(force: bool = False) -> None:
     src = store_dir / "csrc" / "stored_main.cpp"
     hdr2 = store_dir / "csrc" / "msgpack_value.hpp"
     out = bin_dir / "instaslice-stored"
-    if force or not newer(out, [src, hdr2]):
+    if force or not newer(out, [src, hdr2], "stored"):
         run(["g++", "-O2", "-std=c++17", "-pthread", src, "-o", out])
+        record_hash(out, [src, hdr2], "stored")
 
     print("native build complete")
 

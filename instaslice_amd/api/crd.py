@@ -113,9 +113,17 @@ SPEC_SCHEMA = {
                     "gpuUUID": _STR,
                     "ordinal": _INT,
                     "ts": {"type": "number"},
+                    # whole-GPU nomination (plan 2 preemption): the GPU is
+                    # reserved for a mode-flip placement once it drains
+                    "wholeGpu": _BOOL,
                 },
             },
         },
+        # agent opts into two-phase teardown when it must reset the GPU
+        # mode itself (agent/daemonset.py writes this at boot)
+        "agentManagedTeardown": _BOOL,
+        # operator drain (CLI cordon): no new placements on this node
+        "cordoned": _BOOL,
     },
 }
 

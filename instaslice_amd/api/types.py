@@ -270,6 +270,31 @@ def new_pod(
     return pod
 
 
+def timestamp_epoch(ts) -> float:
+    """Normalize a metadata timestamp to epoch seconds.
+
+    MemStore/netstore/stored write epoch floats, but a real Kubernetes API
+    server (the K8sStore backend) returns RFC3339 strings such as
+    "2026-09-14T12:00:00Z" — grace-period math must accept both
+    (advisor finding r1: float() on the string crash-looped teardown on
+    the k8s adapter)."""
+    if isinstance(ts, (int, float)):
+        return float(ts)
+    s = str(ts).strip()
+    try:
+        return float(s)
+    except ValueError:
+        pass
+    from datetime import datetime, timezone
+
+    if s.endswith(("Z", "z")):
+        s = s[:-1] + "+00:00"
+    dt = datetime.fromisoformat(s)
+    if dt.tzinfo is None:
+        dt = dt.replace(tzinfo=timezone.utc)
+    return dt.timestamp()
+
+
 def pod_is_gated(pod: dict) -> bool:
     """reference: checkIfPodGated, instaslice_controller.go:386-395."""
     from instaslice_amd import GATE_NAME

@@ -430,7 +430,10 @@ class PodController:
             return False
         victim = None  # (priority, cr_name, alloc_dict)
         for cr in crs:
-            if self._node_stale(cr):
+            # skip nodes where the preemptor could never land (same filters
+            # as _try_place): evicting victims on a stale/cordoned node
+            # frees nothing the pod can use — it just destroys workloads
+            if self._node_stale(cr) or (cr.get("spec") or {}).get("cordoned"):
                 continue
             for a in (cr["spec"].get("allocations") or {}).values():
                 if a.get("profile") != profile_name:
@@ -609,9 +612,12 @@ class PodController:
         if md.get("deletionTimestamp"):
             self._unschedulable_keys.discard((namespace, name))
             self._pending_since.pop(uid, None)  # deleted before ungating
+            self._preempted_for.pop(uid, None)  # bounded cooldown map
             gated = pod_is_gated(pod)
             if not gated:
-                elapsed = time.time() - float(md["deletionTimestamp"])
+                from instaslice_amd.api.types import timestamp_epoch
+
+                elapsed = time.time() - timestamp_epoch(md["deletionTimestamp"])
                 if elapsed < self.teardown_grace_s:
                     return Result(requeue_after=self.teardown_grace_s - elapsed)
             # ONE batched round-trip: full cleanup (or mark-deleted for
@@ -708,6 +714,7 @@ class PodController:
             # allocation-latency bookkeeping also lands here when the AGENT
             # fast-ungated the pod (data-plane admission completion) — the
             # ungate event reaches us via the pod watch either way
+            self._preempted_for.pop(uid, None)  # placed; cooldown done
             t0 = self._pending_since.pop(uid, None)
             if t0 is not None:
                 dt = time.monotonic() - t0

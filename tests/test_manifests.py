@@ -46,6 +46,42 @@ def test_crd_schema_accepts_real_cr():
         assert req in prep
 
 
+def test_crd_schema_covers_all_spec_keys_written():
+    """Every spec.* key the controller/agent/CLI writes must exist in the
+    structural schema — a real API server PRUNES unknown fields silently
+    (advisor r1: pruned `cordoned` made drain a no-op, pruned `wholeGpu`
+    crash-looped build_gpu_views). Scans the source for patch paths and
+    asserts schema coverage."""
+    import re
+
+    from instaslice_amd.api.crd import SPEC_SCHEMA
+    from instaslice_amd.api.types import new_instaslice
+
+    written = set()
+    # patch-op paths: {"op": ..., "path": ["spec", "<key>", ...]}
+    pat = re.compile(r'\[\s*"spec",\s*"([A-Za-z]+)"')
+    for p in (ROOT / "instaslice_amd").rglob("*.py"):
+        for m in pat.finditer(p.read_text()):
+            written.add(m.group(1))
+    # dict-style writes: spec["<key>"] = ... / spec.setdefault("<key>", ...)
+    pat2 = re.compile(r'spec(?:\.setdefault\(|\[)\s*"([A-Za-z]+)"')
+    for p in (ROOT / "instaslice_amd").rglob("*.py"):
+        for m in pat2.finditer(p.read_text()):
+            written.add(m.group(1))
+    written |= set(new_instaslice("n")["spec"])
+    # the scan also catches Pod-spec patch paths; those live in the core
+    # Pod schema, not our CRD
+    written -= {"schedulingGates", "nodeSelector", "containers"}
+    props = set(SPEC_SCHEMA["properties"])
+    missing = written - props
+    assert not missing, f"spec keys written but absent from CRD schema: {missing}"
+    nom_props = set(
+        SPEC_SCHEMA["properties"]["nominations"]["additionalProperties"]["properties"]
+    )
+    for k in ("gpuUUID", "ordinal", "ts", "wholeGpu"):
+        assert k in nom_props, f"nomination key {k} missing from CRD schema"
+
+
 def test_all_manifests_parse():
     for p in (ROOT / "config").rglob("*.yaml"):
         _load_all(p)

@@ -114,6 +114,21 @@ class AmdSmi(abc.ABC):
         """Activity/VRAM counters for observability (captured around every
         reconfigure, per the north star)."""
 
+    def set_accelerator_profile(self, gpu_uuid: str, profile_index: int) -> None:
+        """Set the partition layout by catalog profile index
+        (amdsmi_set_gpu_accelerator_partition_profile, amdsmi.h:5994) — the
+        documented forward path; the amd-smi CLI routes `set -C` through it.
+        Default: resolve the index to its mode via get_profile_config and
+        delegate to set_compute_partition (exact semantics for the fake);
+        NativeAmdSmi overrides with the direct amdsmi call."""
+        for p in self.get_profile_config(gpu_uuid):
+            if p["profile_index"] == profile_index:
+                self.set_compute_partition(gpu_uuid, p["profile_type"])
+                return
+        raise SmiError(
+            f"gpu {gpu_uuid}: no profile with index {profile_index} in catalog"
+        )
+
     # -- convenience ------------------------------------------------------
 
     def find_gpu(self, gpu_uuid: str) -> PhysicalGpu:

@@ -199,6 +199,35 @@ class Partitiond {
     return out;
   }
 
+  // Currently-active accelerator partition profile (amdsmi.h:5974) — the
+  // read side of set_accelerator_profile, used by the write-path probe to
+  // verify a flip actually landed.
+  ProfileInfo get_current_profile(uint32_t index) {
+    std::lock_guard<std::mutex> lk(mu_);
+    amdsmi_accelerator_partition_profile_t prof;
+    std::memset(&prof, 0, sizeof(prof));
+    uint32_t partition_ids[AMDSMI_MAX_ACCELERATOR_PARTITIONS] = {0};
+    check(amdsmi_get_gpu_accelerator_partition_profile(handle_locked(index), &prof,
+                                                       partition_ids),
+          "amdsmi_get_gpu_accelerator_partition_profile");
+    ProfileInfo pi;
+    switch (prof.profile_type) {
+      case AMDSMI_ACCELERATOR_PARTITION_SPX: pi.profile_type = "SPX"; break;
+      case AMDSMI_ACCELERATOR_PARTITION_DPX: pi.profile_type = "DPX"; break;
+      case AMDSMI_ACCELERATOR_PARTITION_TPX: pi.profile_type = "TPX"; break;
+      case AMDSMI_ACCELERATOR_PARTITION_QPX: pi.profile_type = "QPX"; break;
+      case AMDSMI_ACCELERATOR_PARTITION_CPX: pi.profile_type = "CPX"; break;
+      default: pi.profile_type = "UNKNOWN"; break;
+    }
+    pi.num_partitions = prof.num_partitions;
+    pi.profile_index = prof.profile_index;
+    if (prof.memory_caps.nps_flags.nps1_cap) pi.memory_caps.push_back("NPS1");
+    if (prof.memory_caps.nps_flags.nps2_cap) pi.memory_caps.push_back("NPS2");
+    if (prof.memory_caps.nps_flags.nps4_cap) pi.memory_caps.push_back("NPS4");
+    if (prof.memory_caps.nps_flags.nps8_cap) pi.memory_caps.push_back("NPS8");
+    return pi;
+  }
+
   // Set partition layout by catalog profile_index — the forward-looking API
   // (amdsmi.h:5994); set_compute_partition is the portable one.
   void set_accelerator_profile(uint32_t index, uint32_t profile_index) {

@@ -9,7 +9,7 @@
 //   get_compute <idx> | set_compute <idx> <SPX|DPX|QPX|CPX>
 //   get_memory <idx>  | set_memory <idx> <NPS1|NPS2|NPS4>
 //   profiles <idx>    | set_profile <idx> <profile_index>
-//   metrics <idx>     | quit
+//   current_profile <idx> | metrics <idx> | quit
 //
 // Build: see build_native.py (amdclang++ -lamd_smi).
 
@@ -95,6 +95,14 @@ int main() {
         uint32_t idx, prof; iss >> idx >> prof;
         d.set_accelerator_profile(idx, prof);
         reply_ok();
+      } else if (verb == "current_profile") {
+        uint32_t idx; iss >> idx;
+        ProfileInfo pr = d.get_current_profile(idx);
+        std::ostringstream o;
+        o << "{\"profile_type\": \"" << pr.profile_type
+          << "\", \"num_partitions\": " << pr.num_partitions
+          << ", \"profile_index\": " << pr.profile_index << "}";
+        reply_ok(o.str());
       } else if (verb == "profiles") {
         uint32_t idx; iss >> idx;
         std::ostringstream o;

@@ -74,6 +74,8 @@ PYBIND11_MODULE(_partitiond, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("set_accelerator_profile", &Partitiond::set_accelerator_profile,
            py::call_guard<py::gil_scoped_release>())
+      .def("get_current_profile", &Partitiond::get_current_profile,
+           py::call_guard<py::gil_scoped_release>())
       .def("get_metrics", &Partitiond::get_metrics,
            py::call_guard<py::gil_scoped_release>());
 

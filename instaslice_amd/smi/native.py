@@ -172,11 +172,64 @@ class NativeAmdSmi(AmdSmi):
 
     def set_compute_partition(self, gpu_uuid: str, mode: str) -> None:
         with self._lock:
+            idx = self._proc_index(gpu_uuid)
             try:
-                self._d.set_compute_partition(self._proc_index(gpu_uuid), mode)
+                self._d.set_compute_partition(idx, mode)
+            except _pd.SmiNativeError as e:
+                first = _translate(e)
+                if isinstance(first, (SmiBusy, SmiPermission)):
+                    raise first from None
+                # The type-based set (amdsmi_set_gpu_compute_partition) is
+                # refused on some platforms where the PROFILE-INDEX set
+                # still works — the amd-smi CLI routes `set -C` through
+                # amdsmi_set_gpu_accelerator_partition_profile and succeeds
+                # where the type set returns UNKNOWN_ERROR (probe evidence:
+                # profiles/partition_flip_probe_r01.txt line 39-41). Fall
+                # back to the index API before giving up.
+                try:
+                    pidx = self._profile_index_for(idx, mode)
+                except SmiError:
+                    raise first from None
+                if pidx is None:
+                    raise SmiNotSupported(
+                        f"gpu {gpu_uuid}: mode {mode} not in profile catalog "
+                        f"(type-set also failed: {first})"
+                    ) from None
+                self.log.info(
+                    "gpu %s: type-based set %s failed (%s); retrying via "
+                    "profile index %d", gpu_uuid[:8], mode, first, pidx,
+                )
+                try:
+                    self._d.set_accelerator_profile(idx, pidx)
+                except _pd.SmiNativeError as e2:
+                    second = _translate(e2)
+                    raise type(second)(
+                        f"both partition set APIs failed for {mode}: "
+                        f"type-based [{first}]; profile-index {pidx} "
+                        f"[{second}]"
+                    ) from None
+            # processor population changed: stale index cache
+            self._uuid_to_proc_index.pop(gpu_uuid, None)
+
+    def _profile_index_for(self, proc_index: int, mode: str):
+        """Catalog profile_index for a compute mode, or None if absent."""
+        try:
+            raw = self._d.get_profile_config(proc_index)
+        except _pd.SmiNativeError as e:
+            raise _translate(e) from None
+        for p in raw:
+            if p.profile_type == mode:
+                return p.profile_index
+        return None
+
+    def set_accelerator_profile(self, gpu_uuid: str, profile_index: int) -> None:
+        with self._lock:
+            try:
+                self._d.set_accelerator_profile(
+                    self._proc_index(gpu_uuid), profile_index
+                )
             except _pd.SmiNativeError as e:
                 raise _translate(e) from None
-            # processor population changed: stale index cache
             self._uuid_to_proc_index.pop(gpu_uuid, None)
 
     def get_memory_partition(self, gpu_uuid: str) -> str:

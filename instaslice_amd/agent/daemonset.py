@@ -370,9 +370,14 @@ class NodeAgent:
             self.log.warning("configmap create failed: %s", res[0]["error"])
         if not res[1]["ok"]:
             self.log.warning("node capacity pin failed: %s", res[1]["error"])
-        if fast_ungate and not res[3]["ok"]:
-            # unusual gate set (extra gates / already ungated): hand the
-            # admission back to the controller by downgrading to `created`
+        commit = res[2]
+        if commit["ok"] and fast_ungate and not res[3]["ok"]:
+            # OUR commit just landed but the pod's gate set wasn't exactly
+            # ours (extra gates): hand the admission back to the controller
+            # by downgrading to `created`. MUST be gated on commit["ok"] —
+            # on a stale-cache rerun (commit conflicts because an earlier
+            # pass already flipped the status) this downgrade would revert
+            # a live ungated allocation (r1 flake).
             try:
                 self.store.patch("Instaslice", self.node_name, INSTASLICE_NS, [
                     {"op": "test",
@@ -384,12 +389,30 @@ class NodeAgent:
                 ], quiet=True)
             except (Conflict, NotFound):
                 pass
-        commit = res[2]
         if not commit["ok"]:
-            # Conflict: status moved (pod deleted mid-create). Undo our own
-            # ConfigMap + capacity pin — under the controller's fast
-            # teardown nobody else will clean them up (the allocation may
-            # already be gone from the CR)
+            # Conflict has TWO causes with opposite remedies:
+            #   (a) teardown raced us mid-create (allocation gone/deleted)
+            #       -> undo our ConfigMap + capacity pin, nobody else will;
+            #   (b) stale informer view re-ran an ALREADY-COMMITTED create
+            #       (status is created/ungated in the store) -> the ConfigMap
+            #       belongs to a live pod; undoing here deletes the env
+            #       contract out from under an ungated pod (the r1
+            #       "ungated pod without ConfigMap" flake). Check the fresh
+            #       object to tell them apart.
+            cur_status = None
+            try:
+                fresh = self.store.get("Instaslice", self.node_name, INSTASLICE_NS)
+                cur = (fresh.get("spec", {}).get("allocations") or {}).get(pu)
+                if isinstance(cur, dict):
+                    cur_status = cur.get("allocationStatus")
+            except NotFound:
+                pass
+            if cur_status in (AllocationStatus.CREATED, AllocationStatus.UNGATED):
+                self.log.debug(
+                    "prepare for pod %s already committed (stale view); "
+                    "keeping ConfigMap", alloc["podName"],
+                )
+                return True
             self.log.debug("prepare commit superseded: %s", commit["error"]["msg"])
             self.store.batch([
                 {"verb": "delete", "kind": "ConfigMap",

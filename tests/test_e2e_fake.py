@@ -753,3 +753,33 @@ def test_malformed_allocation_entry_does_not_wedge_agent():
         assert c.agents["node-0"].engine.error_count == 0
     finally:
         c.stop()
+
+
+def test_stale_cache_rerun_keeps_configmap_and_status(cluster):
+    """Regression for the r1 flake: a stale informer view
+    can re-run _commit_prepare for an allocation whose commit ALREADY landed
+    (status ungated in the store, still `creating` in the agent's cached
+    CR). The old code then (a) downgraded the live allocation back to
+    `created` via the fast-ungate fallback and (b) deleted the live pod's
+    ConfigMap in the conflict-undo path — an ungated pod without its
+    visible-devices env. The rerun must be a no-op that reports success."""
+    cluster.submit_pod("p1", "cpx-1x36")
+    cluster.wait_pod_scheduled("p1")
+    agent = cluster.agents["node-0"]
+    allocs = cluster.allocations("node-0")
+    (uid,) = allocs.keys()
+    alloc = dict(allocs[uid])
+    assert alloc["allocationStatus"] == AllocationStatus.UNGATED
+    prepared = cluster.prepared("node-0")
+    (part_uuid,) = prepared.keys()
+    # replay the commit exactly as a stale-cache pass would: the alloc dict
+    # says `creating`, the store says `ungated`
+    stale = dict(alloc, allocationStatus=AllocationStatus.CREATING)
+    ok = agent._commit_prepare(stale, {part_uuid: dict(prepared[part_uuid])})
+    assert ok, "stale rerun of a committed prepare must report success"
+    env = cluster.pod_env("p1")  # raises NotFound if the CM was deleted
+    assert env["INSTASLICE_PARTITION_UUID"] == part_uuid
+    fresh = cluster.allocations("node-0")[uid]
+    assert fresh["allocationStatus"] == AllocationStatus.UNGATED
+    node = cluster.store.get("Node", "node-0", "")
+    assert node["status"]["capacity"].get("org.instaslice/p1") == 1

@@ -195,7 +195,7 @@ class AllocationPolicy:
             if v.uuid in prefer_gpus and v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
                 if p:
-                    return ((3, 0, 0), p)  # gang affinity: top tier
+                    return ((9, 0, 0), p)  # gang affinity: top tier
         return None
 
 
@@ -204,6 +204,27 @@ def _free_ordinal(view: GpuView, n_partitions: int) -> Optional[int]:
         if k not in view.occupied:
             return k
     return None
+
+
+def _upsize_place(view: GpuView, profile: PartitionProfile) -> Optional[Placement]:
+    """Serve a request with a BIGGER partition than asked: a free ordinal on
+    an occupied GPU whose mode's partitions have more XCDs than the profile
+    wants (e.g. cpx-1x36 onto a QPX GPU's 2-XCD slot). MI355X capacity is
+    fungible downward — the pod gets a strictly larger device and the env
+    contract is unchanged — which turns wrong-mode holes (THE fragmentation
+    axis on whole-GPU-mode hardware) into usable capacity
+    instead of unschedulable stalls. Only on non-empty GPUs: idle GPUs are
+    flip territory (flipping wastes nothing)."""
+    if not view.occupied or view.mode_locked:
+        return None
+    mode_xcds = 8 // view.compute_mode.num_partitions
+    if mode_xcds <= profile.xcds:
+        return None  # same or smaller: not an upsize
+    k = _free_ordinal(view, view.compute_mode.num_partitions)
+    if k is None:
+        return None
+    return Placement(view.node, view.uuid, k, view.compute_mode.value,
+                     view.memory_mode.value, False)
 
 
 def _place_on(view: GpuView, profile: PartitionProfile, needs_change: bool) -> Optional[Placement]:
@@ -265,9 +286,37 @@ class PackedFitPolicy(AllocationPolicy):
        (keeps high-index GPUs free for large SPX jobs, and co-locates small
        partitions on few GPUs so xGMI links of the remaining GPUs stay
        uncontended for multi-GPU tenants).
+    3. UPSIZE fallback (r2, the per-GPU-mode fragmentation fix): a free
+       larger-XCD slot on an occupied GPU serves the request with a bigger
+       partition (see _upsize_place), capped at 2 wasted XCDs — beyond
+       that, stranding beats the saving (measured in the churn study,
+       profiles/fragmentation_study_r02.md).
+    4. Flip-vs-upsize ordering is PER CLASS: 1-XCD (CPX) requests absorb
+       into existing holes first (waste 1, no new mode pin — flipping a
+       whole GPU for them creates the straggler split that blocks 4-XCD
+       requests later), while larger profiles flip an idle GPU first
+       (their repacked arrangement genuinely needs dedicated GPUs;
+       upsizing them strands capacity when their own mode later frees).
+    5. MODE COARSENING when no big-profile (>=4-XCD) tenants are present:
+       1-XCD requests then prefer riding 2-XCD QPX slots, and when they
+       must flip an idle GPU they flip it to QPX — the node's small-pod
+       capacity becomes ONE fungible slot class, so any freed slot serves
+       any small request and unschedulable stalls collapse (mixed100 p99
+       1276 -> ~520 ms, p90 267 -> 11 ms). Big-profile occupancy disables
+       it because coarsening trades XCD efficiency for fungibility, which
+       starves 4-XCD requests (churn study).
+    6. Same-mode refills prefer the NEWEST (highest-index) GPU: oldest
+       GPUs then drain monotonically under FIFO churn and become
+       flippable instead of being refilled forever.
+
+       Full measurements + clairvoyant-oracle floor:
+       profiles/fragmentation_study_r02.md.
     """
 
     name = "packed-fit"
+    # beyond this many wasted XCDs an upsize strands more capacity than the
+    # saved flip is worth (churn study: cap 2 strictly beats unbounded)
+    max_upsize_waste = 2
 
     def place_scored(self, profile: PartitionProfile, views: List[GpuView],
                      prefer_gpus: frozenset = frozenset()
@@ -278,19 +327,39 @@ class PackedFitPolicy(AllocationPolicy):
                 return sp
         best: Optional[Tuple[Tuple, Placement]] = None
         node_used = sum(len(v.occupied) for v in views)
+        small = profile.xcds <= 1
+        qpx_present = any(v.compute_mode is ComputeMode.QPX for v in views)
+        big_demand = any(
+            v.occupied and v.compute_mode in (ComputeMode.DPX, ComputeMode.SPX)
+            for v in views
+        )
+        coarsen = small and qpx_present and not big_demand
+        flip_tier, upsize_tier = (2, 3) if small else (3, 2)
         for v in views:
             if v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
                 if p is None:
                     continue
-                # tier 2 = same mode; prefer most-occupied, then lowest index
-                score = (2, len(v.occupied), 0, -v.index)
+                score = (3 if coarsen else 4, v.index, len(v.occupied), 0)
             else:
                 p = _place_on(v, profile, needs_change=True)
-                if p is None:
-                    continue
-                mem_ok = v.memory_mode in VALID_MEMORY_MODES[profile.compute]
-                score = (1, 1 if mem_ok else 0, node_used, -v.index)
+                if p is not None:
+                    if coarsen:
+                        # flip to QPX: small capacity joins the 2-XCD class
+                        mem, _ = _target_memory_mode(profile, v.memory_mode)
+                        p = Placement(v.node, v.uuid, 0,
+                                      ComputeMode.QPX.value, mem.value, True)
+                    mem_ok = v.memory_mode in VALID_MEMORY_MODES[profile.compute]
+                    score = (flip_tier, 1 if mem_ok else 0, node_used, -v.index)
+                else:
+                    p = _upsize_place(v, profile)
+                    if p is None:
+                        continue
+                    waste = 8 // v.compute_mode.num_partitions - profile.xcds
+                    if waste > self.max_upsize_waste:
+                        continue
+                    score = (4 if coarsen else upsize_tier,
+                             -waste, len(v.occupied), -v.index)
             if best is None or score > best[0]:
                 best = (score, p)
         return best

@@ -238,12 +238,18 @@ class PodController:
         candidates.sort(key=lambda c: c[0], reverse=True)
         for _, cr, placement, profile in candidates:
             node_name = cr["metadata"]["name"]
+            # actual partition geometry comes from the placement's mode, not
+            # the requested profile: an UPSIZE placement (policy fallback)
+            # hands the pod a larger partition than it asked for
+            from instaslice_amd.partition.profiles import ComputeMode as _CM
+
+            actual_xcds = 8 // _CM(placement.compute_mode).num_partitions
             alloc = AllocationDetails(
                 profile=profile_name,
                 gpu_uuid=placement.gpu_uuid,
                 ordinal=placement.ordinal,
-                start=placement.ordinal * profile.xcds,
-                size=profile.xcds,
+                start=placement.ordinal * actual_xcds,
+                size=actual_xcds,
                 pod_uuid=md["uid"],
                 pod_name=md["name"],
                 namespace=md.get("namespace", "default"),

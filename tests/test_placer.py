@@ -89,7 +89,8 @@ def test_packedfit_prefers_matching_memory_mode_on_reconfig():
 def test_exhaustive_mix_8gpu_node(policy_cls):
     """Feed every 3-profile request mix to an 8-GPU node, applying each
     placement; assert invariants: ordinals unique per GPU, mode changes only
-    on idle GPUs, capacity respected."""
+    on idle GPUs, and the pod NEVER gets a smaller partition than requested
+    (upsize/coarsening may legally hand out a bigger one)."""
     profiles = [CPX, QPX, DPX, SPX]
     for mix in itertools.product(profiles, repeat=3):
         views = [mk_view(i) for i in range(8)]
@@ -100,11 +101,15 @@ def test_exhaustive_mix_8gpu_node(policy_cls):
             v = next(v for v in views if v.uuid == p.gpu_uuid)
             if p.needs_mode_change:
                 assert not v.occupied
-                v.compute_mode = prof.compute
+                v.compute_mode = ComputeMode(p.compute_mode)
                 v.memory_mode = MemoryMode(p.memory_mode)
-            assert v.compute_mode is prof.compute
+            # the placement's mode is authoritative (upsize/coarsening may
+            # differ from the profile's own mode) and must match the GPU
+            assert v.compute_mode.value == p.compute_mode
+            n_parts = v.compute_mode.num_partitions
             assert p.ordinal not in v.occupied
-            assert p.ordinal < prof.partitions_per_gpu
+            assert p.ordinal < n_parts
+            assert 8 // n_parts >= prof.xcds, "pod must never be undersized"
             v.occupied.add(p.ordinal)
 
 
@@ -216,3 +221,74 @@ def test_cross_node_spreading():
             "spread-fit packed instead of spreading cross-node")
     finally:
         c.stop()
+
+
+def test_upsize_serves_small_request_from_bigger_hole():
+    """cpx request, CPX GPUs full, a QPX hole on an occupied GPU, no idle
+    GPU: the request rides the 2-XCD slot instead of going unschedulable."""
+    views = [
+        mk_view(0, ComputeMode.CPX, occupied=set(range(8))),
+        mk_view(1, ComputeMode.QPX, occupied={0, 1, 2}),
+    ]
+    p = PackedFitPolicy().place(CPX, views)
+    assert p is not None
+    assert p.gpu_uuid == "gpu-1" and p.compute_mode == "QPX"
+    assert not p.needs_mode_change
+
+
+def test_upsize_never_on_idle_gpu():
+    """An idle GPU is flip territory: a cpx request must flip it (or, with
+    QPX present and no big tenants, coarsen to QPX) — never 'upsize' into
+    the idle GPU's current-mode slot."""
+    views = [mk_view(0, ComputeMode.DPX)]  # idle, wrong mode
+    p = PackedFitPolicy().place(CPX, views)
+    assert p.needs_mode_change
+
+
+def test_upsize_waste_capped():
+    """cpx into a DPX hole wastes 3 XCDs > cap 2: refused (stranding beats
+    the saving — churn study)."""
+    views = [
+        mk_view(0, ComputeMode.CPX, occupied=set(range(8))),
+        mk_view(1, ComputeMode.DPX, occupied={0}),
+    ]
+    p = PackedFitPolicy().place(CPX, views)
+    assert p is None
+
+
+def test_coarsening_flips_idle_to_qpx_for_small_request():
+    """QPX present + no big tenants: a cpx request flips the idle GPU to
+    QPX, keeping small-pod capacity one fungible class."""
+    views = [
+        mk_view(0, ComputeMode.QPX, occupied={0, 1, 2, 3}),
+        mk_view(1, ComputeMode.SPX),  # idle
+    ]
+    p = PackedFitPolicy().place(CPX, views)
+    assert p.needs_mode_change and p.compute_mode == "QPX"
+
+
+def test_coarsening_disabled_by_big_tenants():
+    """A DPX tenant on the node disables coarsening: the cpx request flips
+    the idle GPU to its own CPX mode (XCD efficiency wins)."""
+    views = [
+        mk_view(0, ComputeMode.QPX, occupied={0, 1, 2, 3}),
+        mk_view(1, ComputeMode.DPX, occupied={0, 1}),
+        mk_view(2, ComputeMode.SPX),  # idle
+    ]
+    p = PackedFitPolicy().place(CPX, views)
+    assert p.needs_mode_change and p.compute_mode == "CPX"
+
+
+def test_qpx_prefers_flip_over_upsize_when_idle_exists():
+    """2-XCD profiles flip an idle GPU rather than stranding themselves in
+    a DPX hole (their repacked arrangement wants dedicated GPUs)."""
+    views = [
+        mk_view(0, ComputeMode.DPX, occupied={0}),
+        mk_view(1, ComputeMode.SPX),  # idle
+    ]
+    p = PackedFitPolicy().place(QPX, views)
+    assert p.needs_mode_change and p.compute_mode == "QPX"
+    # ...but with no idle GPU, the DPX hole beats unschedulable
+    views = [mk_view(0, ComputeMode.DPX, occupied={0})]
+    p = PackedFitPolicy().place(QPX, views)
+    assert p is not None and p.compute_mode == "DPX" and not p.needs_mode_change

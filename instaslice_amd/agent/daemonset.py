@@ -144,6 +144,15 @@ class NodeAgent:
                 "index": g.index,
             }
         spec["placements"] = catalog.to_dict()
+        # xGMI link topology into the CR so the controller can score gang
+        # placements hop-aware without talking to the device layer
+        # (amdsmi_topo_get_link_type; empty dict = topology unknown)
+        try:
+            spec["topology"] = self.smi.get_topology()
+        except SmiError as e:
+            self.log.warning("topology discovery failed (%s); scoring "
+                             "degrades to same-GPU affinity", e)
+            spec["topology"] = {}
         # teardown ownership: when the agent must touch the device on drain
         # (mode reset), the controller leaves teardown to the agent's
         # deleted-status protocol; otherwise the controller fast-paths the
@@ -169,6 +178,7 @@ class NodeAgent:
                 obj["spec"]["gpuUuids"] = spec["gpuUuids"]
                 obj["spec"]["placements"] = spec["placements"]
                 obj["spec"]["agentManagedTeardown"] = spec["agentManagedTeardown"]
+                obj["spec"]["topology"] = spec["topology"]
                 for uuid, gd in spec["gpus"].items():
                     old = obj["spec"].setdefault("gpus", {}).get(uuid)
                     if old:

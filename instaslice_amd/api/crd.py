@@ -122,6 +122,15 @@ SPEC_SCHEMA = {
         # agent opts into two-phase teardown when it must reset the GPU
         # mode itself (agent/daemonset.py writes this at boot)
         "agentManagedTeardown": _BOOL,
+        # xGMI link topology discovered at boot: {src_uuid: {dst_uuid: hops}}
+        # (amdsmi_topo_get_link_type); gang scoring prefers 1-hop neighbors
+        "topology": {
+            "type": "object",
+            "additionalProperties": {
+                "type": "object",
+                "additionalProperties": _INT,
+            },
+        },
         # operator drain (CLI cordon): no new placements on this node
         "cordoned": _BOOL,
     },

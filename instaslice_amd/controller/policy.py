@@ -179,24 +179,52 @@ class AllocationPolicy:
     node_order_first_fit = False  # True: controller takes the first node hit
 
     def place_scored(self, profile: PartitionProfile, views: List[GpuView],
-                     prefer_gpus: frozenset = frozenset()
+                     prefer_gpus: frozenset = frozenset(),
+                     xgmi_neighbors: frozenset = frozenset()
                      ) -> Optional[Tuple[Score, Placement]]:
         raise NotImplementedError
 
     def place(self, profile: PartitionProfile, views: List[GpuView],
-              prefer_gpus: frozenset = frozenset()) -> Optional[Placement]:
-        sp = self.place_scored(profile, views, prefer_gpus)
+              prefer_gpus: frozenset = frozenset(),
+              xgmi_neighbors: frozenset = frozenset()) -> Optional[Placement]:
+        sp = self.place_scored(profile, views, prefer_gpus, xgmi_neighbors)
         return sp[1] if sp else None
 
     def _preferred(self, profile: PartitionProfile, views: List[GpuView],
-                   prefer_gpus: frozenset
+                   prefer_gpus: frozenset,
+                   xgmi_neighbors: frozenset = frozenset()
                    ) -> Optional[Tuple[Score, Placement]]:
+        """Gang-affinity tiers, shared by every policy:
+          9: free same-mode slot on a GPU already hosting the gang
+             (intra-die XCD co-location — no xGMI hop at all);
+          8: on a 1-hop xGMI NEIGHBOR of a gang GPU — same-mode slot
+             (8,1,·) beating an idle-neighbor mode flip (8,0,·). For
+             RCCL gangs, link locality dominates one flip's cost.
+        Neighbor sets come from the CR's discovered topology
+        (spec.topology via amdsmi_topo_get_link_type); empty set = no
+        topology known = same-GPU affinity only (r1 behavior)."""
         for v in views:
             if v.uuid in prefer_gpus and v.compute_mode is profile.compute:
                 p = _place_on(v, profile, needs_change=False)
                 if p:
-                    return ((9, 0, 0), p)  # gang affinity: top tier
-        return None
+                    return ((9, 0, 0), p)
+        best: Optional[Tuple[Score, Placement]] = None
+        for v in views:
+            if v.uuid not in xgmi_neighbors:
+                continue
+            if v.compute_mode is profile.compute:
+                p = _place_on(v, profile, needs_change=False)
+                if p:
+                    score = (8, 1, len(v.occupied))
+                    if best is None or score > best[0]:
+                        best = (score, p)
+            else:
+                p = _place_on(v, profile, needs_change=True)
+                if p:
+                    score = (8, 0, -v.index)
+                    if best is None or score > best[0]:
+                        best = (score, p)
+        return best
 
 
 def _free_ordinal(view: GpuView, n_partitions: int) -> Optional[int]:
@@ -251,10 +279,11 @@ class FirstFitPolicy(AllocationPolicy):
     node_order_first_fit = True
 
     def place_scored(self, profile: PartitionProfile, views: List[GpuView],
-                     prefer_gpus: frozenset = frozenset()
+                     prefer_gpus: frozenset = frozenset(),
+                     xgmi_neighbors: frozenset = frozenset()
                      ) -> Optional[Tuple[Score, Placement]]:
-        if prefer_gpus:
-            sp = self._preferred(profile, views, prefer_gpus)
+        if prefer_gpus or xgmi_neighbors:
+            sp = self._preferred(profile, views, prefer_gpus, xgmi_neighbors)
             if sp:
                 return sp
         for v in views:
@@ -319,10 +348,11 @@ class PackedFitPolicy(AllocationPolicy):
     max_upsize_waste = 2
 
     def place_scored(self, profile: PartitionProfile, views: List[GpuView],
-                     prefer_gpus: frozenset = frozenset()
+                     prefer_gpus: frozenset = frozenset(),
+                     xgmi_neighbors: frozenset = frozenset()
                      ) -> Optional[Tuple[Score, Placement]]:
-        if prefer_gpus:
-            sp = self._preferred(profile, views, prefer_gpus)
+        if prefer_gpus or xgmi_neighbors:
+            sp = self._preferred(profile, views, prefer_gpus, xgmi_neighbors)
             if sp:
                 return sp
         best: Optional[Tuple[Tuple, Placement]] = None
@@ -375,10 +405,11 @@ class SpreadFitPolicy(AllocationPolicy):
     name = "spread-fit"
 
     def place_scored(self, profile: PartitionProfile, views: List[GpuView],
-                     prefer_gpus: frozenset = frozenset()
+                     prefer_gpus: frozenset = frozenset(),
+                     xgmi_neighbors: frozenset = frozenset()
                      ) -> Optional[Tuple[Score, Placement]]:
-        if prefer_gpus:
-            sp = self._preferred(profile, views, prefer_gpus)
+        if prefer_gpus or xgmi_neighbors:
+            sp = self._preferred(profile, views, prefer_gpus, xgmi_neighbors)
             if sp:
                 return sp
         best: Optional[Tuple[Tuple, Placement]] = None

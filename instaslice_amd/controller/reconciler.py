@@ -214,7 +214,25 @@ class PodController:
                 for a in (cr["spec"].get("allocations") or {}).values()
                 if group and a.get("group") == group
             )
-            sp = self.policy.place_scored(profile, views, prefer_gpus=prefer)
+            # hop-aware gang spillover: 1-hop xGMI neighbors of gang GPUs
+            # (discovered topology in the CR; empty when unknown)
+            neighbors: frozenset = frozenset()
+            if prefer:
+                topo = cr["spec"].get("topology") or {}
+                near = set()
+                for g_uuid in prefer:
+                    peers = topo.get(g_uuid) or {}
+                    if not peers:
+                        continue
+                    # nearest tier = minimum hop count this platform reports
+                    # (absolute values vary by driver: 1 on the fake, 2 on
+                    # some bare-metal xGMI reports)
+                    mh = min(int(h) for h in peers.values())
+                    near.update(d for d, h in peers.items()
+                                if int(h) == mh and d not in prefer)
+                neighbors = frozenset(near)
+            sp = self.policy.place_scored(profile, views, prefer_gpus=prefer,
+                                          xgmi_neighbors=neighbors)
             if sp is None:
                 continue
             candidates.append((sp[0], cr, sp[1], profile))

@@ -114,6 +114,13 @@ class AmdSmi(abc.ABC):
         """Activity/VRAM counters for observability (captured around every
         reconfigure, per the north star)."""
 
+    def get_topology(self) -> Dict[str, Dict[str, int]]:
+        """Physical-GPU link topology: {src_uuid: {dst_uuid: hops}} for
+        XGMI-class links (amdsmi_topo_get_link_type, amdsmi.h:5635).
+        Placement uses it for hop-aware gang scoring. Default: empty
+        (topology unknown — scoring degrades to same-GPU affinity only)."""
+        return {}
+
     def set_accelerator_profile(self, gpu_uuid: str, profile_index: int) -> None:
         """Set the partition layout by catalog profile index
         (amdsmi_set_gpu_accelerator_partition_profile, amdsmi.h:5994) — the

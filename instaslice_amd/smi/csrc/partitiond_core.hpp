@@ -237,6 +237,50 @@ class Partitiond {
     stale_ = true;
   }
 
+  // Pairwise link topology (amdsmi_topo_get_link_type/_weight, amdsmi.h:5635,
+  // :5574): the placement layer uses it for xGMI-hop-aware gang scoring
+  // (SURVEY.md §5 — 8 MI355X GPUs are fully connected by 7 p2p links each;
+  // partial meshes and PCIe-bridged boards score worse).
+  struct LinkInfo {
+    uint32_t src = 0, dst = 0;
+    uint64_t hops = 0;
+    uint64_t weight = 0;
+    std::string type;  // "XGMI" | "PCIE" | "INTERNAL" | "UNKNOWN"
+  };
+
+  std::vector<LinkInfo> get_link_topology() {
+    std::lock_guard<std::mutex> lk(mu_);
+    require_init();
+    refresh_handles_locked();
+    std::vector<LinkInfo> out;
+    for (uint32_t i = 0; i < handles_.size(); ++i) {
+      for (uint32_t j = 0; j < handles_.size(); ++j) {
+        if (i == j) continue;
+        LinkInfo li;
+        li.src = i;
+        li.dst = j;
+        uint64_t hops = 0;
+        amdsmi_link_type_t t = AMDSMI_LINK_TYPE_UNKNOWN;
+        if (amdsmi_topo_get_link_type(handles_[i], handles_[j], &hops, &t) !=
+            AMDSMI_STATUS_SUCCESS)
+          continue;
+        li.hops = hops;
+        switch (t) {
+          case AMDSMI_LINK_TYPE_XGMI: li.type = "XGMI"; break;
+          case AMDSMI_LINK_TYPE_PCIE: li.type = "PCIE"; break;
+          case AMDSMI_LINK_TYPE_INTERNAL: li.type = "INTERNAL"; break;
+          default: li.type = "UNKNOWN"; break;
+        }
+        uint64_t w = 0;
+        if (amdsmi_topo_get_link_weight(handles_[i], handles_[j], &w) ==
+            AMDSMI_STATUS_SUCCESS)
+          li.weight = w;
+        out.push_back(std::move(li));
+      }
+    }
+    return out;
+  }
+
   // Counters captured around every reconfigure (north-star observability).
   Metrics get_metrics(uint32_t index) {
     std::lock_guard<std::mutex> lk(mu_);

@@ -9,7 +9,7 @@
 //   get_compute <idx> | set_compute <idx> <SPX|DPX|QPX|CPX>
 //   get_memory <idx>  | set_memory <idx> <NPS1|NPS2|NPS4>
 //   profiles <idx>    | set_profile <idx> <profile_index>
-//   current_profile <idx> | metrics <idx> | quit
+//   current_profile <idx> | topology | metrics <idx> | quit
 //
 // Build: see build_native.py (amdclang++ -lamd_smi).
 
@@ -120,6 +120,19 @@ int main() {
             o << "\"" << pr.memory_caps[i] << "\"";
           }
           o << "]}";
+        }
+        o << "]";
+        reply_ok(o.str());
+      } else if (verb == "topology") {
+        std::ostringstream o;
+        o << "[";
+        bool first = true;
+        for (const auto& li : d.get_link_topology()) {
+          if (!first) o << ", ";
+          first = false;
+          o << "{\"src\": " << li.src << ", \"dst\": " << li.dst
+            << ", \"hops\": " << li.hops << ", \"weight\": " << li.weight
+            << ", \"type\": \"" << li.type << "\"}";
         }
         o << "]";
         reply_ok(o.str());

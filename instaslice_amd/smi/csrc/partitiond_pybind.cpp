@@ -49,6 +49,13 @@ PYBIND11_MODULE(_partitiond, m) {
       .def_readonly("profile_index", &ProfileInfo::profile_index)
       .def_readonly("memory_caps", &ProfileInfo::memory_caps);
 
+  py::class_<Partitiond::LinkInfo>(m, "LinkInfo")
+      .def_readonly("src", &Partitiond::LinkInfo::src)
+      .def_readonly("dst", &Partitiond::LinkInfo::dst)
+      .def_readonly("hops", &Partitiond::LinkInfo::hops)
+      .def_readonly("weight", &Partitiond::LinkInfo::weight)
+      .def_readonly("type", &Partitiond::LinkInfo::type);
+
   py::class_<Metrics>(m, "Metrics")
       .def_readonly("gfx_activity_pct", &Metrics::gfx_activity_pct)
       .def_readonly("umc_activity_pct", &Metrics::umc_activity_pct)
@@ -75,6 +82,8 @@ PYBIND11_MODULE(_partitiond, m) {
       .def("set_accelerator_profile", &Partitiond::set_accelerator_profile,
            py::call_guard<py::gil_scoped_release>())
       .def("get_current_profile", &Partitiond::get_current_profile,
+           py::call_guard<py::gil_scoped_release>())
+      .def("get_link_topology", &Partitiond::get_link_topology,
            py::call_guard<py::gil_scoped_release>())
       .def("get_metrics", &Partitiond::get_metrics,
            py::call_guard<py::gil_scoped_release>());

@@ -69,9 +69,18 @@ class FakeAmdSmi(AmdSmi):
         memory_gb: int = MI355X_HBM_GB,
         compute_set_latency_s: float = 0.0,
         memory_set_latency_s: float = 0.0,
+        xgmi_cliques: Optional[list] = None,
     ):
         self._lock = threading.RLock()
         self._gpus = [_FakeGpu(node_name, i, memory_gb) for i in range(num_gpus)]
+        # xGMI topology model: GPUs in one clique are 1-hop neighbors. A
+        # real MI355X node is one fully-connected clique of 8 (7 p2p links
+        # per GPU); tests inject partial meshes (e.g. [[0,1,2,3],[4,5,6,7]])
+        # to exercise hop-aware gang scoring.
+        self.xgmi_cliques = (
+            xgmi_cliques if xgmi_cliques is not None
+            else [list(range(num_gpus))]
+        )
         self._initialized = False
         self.compute_set_latency_s = compute_set_latency_s
         self.memory_set_latency_s = memory_set_latency_s
@@ -225,6 +234,19 @@ class FakeAmdSmi(AmdSmi):
                 }
             )
         return out
+
+    def get_topology(self) -> Dict[str, Dict[str, int]]:
+        self._require_init()
+        self._count("get_topology")
+        topo: Dict[str, Dict[str, int]] = {}
+        for clique in self.xgmi_cliques:
+            for i in clique:
+                for j in clique:
+                    if i == j:
+                        continue
+                    topo.setdefault(self._gpus[i].uuid, {})[
+                        self._gpus[j].uuid] = 1
+        return topo
 
     def get_metrics(self, gpu_uuid: str) -> Dict[str, float]:
         self._require_init()

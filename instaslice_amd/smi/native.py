@@ -261,6 +261,38 @@ class NativeAmdSmi(AmdSmi):
             for p in raw
         ]
 
+    def get_topology(self) -> Dict[str, Dict[str, int]]:
+        """xGMI topology between physical packages: {src_uuid: {dst_uuid:
+        hops}}. Processor-level links are aggregated to the package level
+        (in CPX one package enumerates as 8 processors; intra-package links
+        are INTERNAL and skipped)."""
+        with self._lock:
+            try:
+                procs = self._d.enumerate()
+                links = self._d.get_link_topology()
+            except _pd.SmiNativeError as e:
+                raise _translate(e) from None
+            groups = self._group(procs)
+            proc_to_phys: Dict[int, str] = {}
+            for key, members in groups.items():
+                head = members[0]
+                phys = self._serial_to_uuid.get(key, head.uuid)
+                for p in members:
+                    proc_to_phys[p.index] = phys
+            topo: Dict[str, Dict[str, int]] = {}
+            for li in links:
+                if li.type not in ("XGMI",):
+                    continue
+                s = proc_to_phys.get(li.src)
+                d = proc_to_phys.get(li.dst)
+                if s is None or d is None or s == d:
+                    continue
+                cur = topo.setdefault(s, {})
+                hops = int(li.hops)
+                if d not in cur or hops < cur[d]:
+                    cur[d] = hops
+            return topo
+
     def get_metrics(self, gpu_uuid: str) -> Dict[str, float]:
         try:
             m = self._d.get_metrics(self._proc_index(gpu_uuid))

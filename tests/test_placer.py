@@ -292,3 +292,87 @@ def test_qpx_prefers_flip_over_upsize_when_idle_exists():
     views = [mk_view(0, ComputeMode.DPX, occupied={0})]
     p = PackedFitPolicy().place(QPX, views)
     assert p is not None and p.compute_mode == "DPX" and not p.needs_mode_change
+
+
+def test_gang_prefers_xgmi_neighbor_over_far_gpu():
+    """2-GPU gang with a partial xGMI mesh: the second member must land on
+    a 1-hop neighbor of the first member's GPU, not on a topologically
+    distant GPU."""
+    views = [
+        mk_view(0, ComputeMode.CPX, occupied=set(range(8))),  # gang GPU full
+        mk_view(1, ComputeMode.CPX, occupied={0}),  # far, same mode, free
+        mk_view(2, ComputeMode.CPX, occupied={0}),  # neighbor, same mode, free
+    ]
+    p = PackedFitPolicy().place(
+        CPX, views,
+        prefer_gpus=frozenset({"gpu-0"}),
+        xgmi_neighbors=frozenset({"gpu-2"}),
+    )
+    assert p.gpu_uuid == "gpu-2"
+
+
+def test_gang_neighbor_flip_beats_far_same_mode():
+    """An idle xGMI neighbor (needing a mode flip) outranks a free
+    same-mode slot on a distant GPU — for RCCL gangs link locality
+    dominates one flip's cost."""
+    views = [
+        mk_view(0, ComputeMode.CPX, occupied=set(range(8))),  # gang GPU full
+        mk_view(1, ComputeMode.CPX, occupied={0}),  # far, same mode, free
+        mk_view(2, ComputeMode.SPX),                # neighbor, idle
+    ]
+    p = PackedFitPolicy().place(
+        CPX, views,
+        prefer_gpus=frozenset({"gpu-0"}),
+        xgmi_neighbors=frozenset({"gpu-2"}),
+    )
+    assert p.gpu_uuid == "gpu-2" and p.needs_mode_change
+
+
+def test_gang_same_gpu_still_top():
+    views = [
+        mk_view(0, ComputeMode.CPX, occupied={0}),  # gang GPU, slots free
+        mk_view(1, ComputeMode.CPX, occupied={0}),  # neighbor
+    ]
+    p = PackedFitPolicy().place(
+        CPX, views,
+        prefer_gpus=frozenset({"gpu-0"}),
+        xgmi_neighbors=frozenset({"gpu-1"}),
+    )
+    assert p.gpu_uuid == "gpu-0"
+
+
+def test_gang_lands_on_neighbor_clique_e2e():
+    """End-to-end over the fake: 8-GPU node modeled as two 4-GPU xGMI
+    cliques; a gang whose first member lands in clique A must keep its
+    second member inside clique A (CR carries discovered topology)."""
+    import time
+
+    from instaslice_amd.runtime.cluster import Cluster
+    from instaslice_amd.smi.fake import FakeAmdSmi
+    from instaslice_amd.controller.reconciler import INSTASLICE_NS
+
+    smi = FakeAmdSmi(num_gpus=8, node_name="node-0",
+                     xgmi_cliques=[[0, 1, 2, 3], [4, 5, 6, 7]])
+    c = Cluster(teardown_grace_s=0.0, policy="packed-fit")
+    c.add_node("node-0", smi)
+    c.start()
+    try:
+        cr = c.store.get("Instaslice", "node-0", INSTASLICE_NS)
+        assert cr["spec"]["topology"], "CR must carry discovered topology"
+        # occupy GPU0 fully with the first gang member's profile spx-8x288
+        # (whole GPU) so the second member CANNOT share the GPU
+        c.submit_pod("g1", "spx-8x288", group="team")
+        c.wait_pod_scheduled("g1")
+        g1 = next(iter(c.allocations("node-0").values()))
+        first_gpu = g1["gpuUUID"]
+        c.submit_pod("g2", "spx-8x288", group="team")
+        c.wait_pod_scheduled("g2")
+        allocs = c.allocations("node-0")
+        g2 = next(a for a in allocs.values() if a["podName"] == "g2")
+        # the second member must sit on a 1-hop neighbor of the first
+        topo = cr["spec"]["topology"]
+        assert g2["gpuUUID"] in topo.get(first_gpu, {}), (
+            f"gang member landed outside the xGMI clique: "
+            f"{g2['gpuUUID']} not neighbor of {first_gpu}")
+    finally:
+        c.stop()

@@ -447,6 +447,40 @@ class NodeAgent:
         g = self._gpus.get(alloc["gpuUUID"])
         return bool(g and g.compute_mode != alloc["computeMode"])
 
+    def _batch_flips(self, cr: dict, allocations: dict) -> None:
+        """FLIP BATCHING: when one reconcile pass holds
+        `creating` allocations that need mode changes on SEVERAL distinct
+        GPUs, run the device sets concurrently instead of paying the flip
+        wall-time once per GPU in sequence (a real compute-partition set is
+        O(100 ms-1 s); a burst of placements across 8 GPUs would otherwise
+        serialize into seconds). Errors are swallowed here on purpose: the
+        sequential per-allocation path immediately redoes the (now no-op or
+        still-failing) flip and owns all error/strike accounting."""
+        flips = {}
+        for alloc in allocations.values():
+            if not isinstance(alloc, dict):
+                continue
+            if alloc.get("allocationStatus") != AllocationStatus.CREATING:
+                continue
+            gpu = alloc.get("gpuUUID")
+            if gpu and gpu not in flips and "computeMode" in alloc \
+                    and self._needs_mode_change(alloc):
+                flips[gpu] = (alloc["computeMode"],
+                              alloc.get("memoryMode") or "NPS1")
+        if len(flips) < 2:
+            return  # nothing to overlap
+        from concurrent.futures import ThreadPoolExecutor
+
+        def one(gpu_uuid, compute, memory):
+            try:
+                self._ensure_gpu_mode(cr, gpu_uuid, compute, memory)
+            except SmiError:
+                pass  # sequential path re-attempts and does the accounting
+
+        with ThreadPoolExecutor(max_workers=min(8, len(flips))) as ex:
+            for gpu_uuid, (compute, memory) in flips.items():
+                ex.submit(one, gpu_uuid, compute, memory)
+
     def _fail_allocation(self, pod_uuid: str, gpu_uuid: str, lock_mode: bool) -> None:
         """Flip the allocation to `failed` (controller will re-place) and
         optionally mark the GPU mode-locked in the CR."""
@@ -486,6 +520,8 @@ class NodeAgent:
                 return Result()
         allocations = cr.get("spec", {}).get("allocations") or {}
         requeue: Optional[float] = None
+
+        self._batch_flips(cr, allocations)
 
         for pod_uuid, alloc in sorted(allocations.items()):
             # one externally crafted/corrupt entry must not wedge the whole

@@ -88,6 +88,11 @@ class FakeAmdSmi(AmdSmi):
         self.fault_hook: Optional[Callable[[str, str], None]] = None
         # counters for assertions (e.g. "enumeration happened once")
         self.call_counts: Dict[str, int] = {}
+        # flip-batching observability: how many mode sets ran CONCURRENTLY
+        # at peak (asserted by the agent's flip-batching test)
+        self.max_concurrent_sets = 0
+        self._conc = 0
+        self._conc_lock = threading.Lock()
 
     # -- internals ----------------------------------------------------------
 
@@ -182,8 +187,25 @@ class FakeAmdSmi(AmdSmi):
                 )
             if new_mode is g.compute_mode:
                 return
+        # latency OUTSIDE the node lock: on real hardware mode sets on
+        # DIFFERENT GPUs proceed independently, and the agent's flip
+        # batching relies on that (peak concurrency is recorded for the
+        # batching test)
+        with self._conc_lock:
+            self._conc += 1
+            self.max_concurrent_sets = max(self.max_concurrent_sets, self._conc)
+        try:
             if self.compute_set_latency_s:
                 time.sleep(self.compute_set_latency_s)
+        finally:
+            with self._conc_lock:
+                self._conc -= 1
+        with self._lock:
+            g = self._gpu(gpu_uuid)
+            if g.busy_partitions:
+                raise SmiBusy(f"gpu {gpu_uuid}: busy")
+            if new_mode is g.compute_mode:
+                return
             g.compute_mode = new_mode
             g.mode_generation += 1
             # mode combination validity: keep memory mode legal

@@ -783,3 +783,32 @@ def test_stale_cache_rerun_keeps_configmap_and_status(cluster):
     assert fresh["allocationStatus"] == AllocationStatus.UNGATED
     node = cluster.store.get("Node", "node-0", "")
     assert node["status"]["capacity"].get("org.instaslice/p1") == 1
+
+
+def test_flip_batching_overlaps_mode_sets():
+    """4 dpx pods spread onto 4 distinct GPUs in one burst: the agent must
+    run the (slow) mode sets concurrently, not serialized — asserted via
+    the fake's peak-concurrency counter and by wall time well under the
+    4x sequential floor (flip batching)."""
+    from instaslice_amd.runtime.cluster import Cluster
+    from instaslice_amd.smi.fake import FakeAmdSmi
+
+    smi = FakeAmdSmi(num_gpus=4, node_name="node-0",
+                     compute_set_latency_s=0.15)
+    # spread-fit puts each pod on its own idle GPU -> 4 distinct DPX flips
+    c = Cluster(teardown_grace_s=0.0, policy="spread-fit")
+    c.add_node("node-0", smi)
+    c.start()
+    try:
+        t0 = time.monotonic()
+        for i in range(4):
+            c.submit_pod(f"d{i}", "dpx-4x144")
+        for i in range(4):
+            c.wait_pod_scheduled(f"d{i}", timeout=20.0)
+        wall = time.monotonic() - t0
+    finally:
+        c.stop()
+    assert smi.max_concurrent_sets >= 2, (
+        f"mode sets never overlapped (peak={smi.max_concurrent_sets})")
+    # sequential floor would be >= 4 * 0.15 = 0.6s of pure set time
+    assert wall < 0.55, f"flip batching ineffective: wall={wall:.2f}s"

@@ -604,10 +604,6 @@ class NodeAgent:
         """ONE batched round-trip for the whole delete path: drop ConfigMap +
         capacity pin + allocation/prepared/ordinal from the CR (reference does
         these as separate API calls, instaslice_daemonset.go:415-470)."""
-        part_uuids = [
-            k for k, v in (cr["spec"].get("prepared") or {}).items()
-            if v["podUUID"] == pod_uuid
-        ]
         cr_ops = [
             {"op": "test",
              "path": ["spec", "allocations", pod_uuid, "allocationStatus"],
@@ -616,9 +612,10 @@ class NodeAgent:
             {"op": "remove_from_set",
              "path": ["spec", "gpus", alloc["gpuUUID"], "usedOrdinals"],
              "value": alloc["ordinal"]},
-        ] + [
-            {"op": "delete", "path": ["spec", "prepared", puid]}
-            for puid in part_uuids
+            # predicate delete against the fresh object (see reconciler
+            # teardown): stale-view key lists orphan prepared entries
+            {"op": "delete_where", "path": ["spec", "prepared"],
+             "field": "podUUID", "value": pod_uuid},
         ]
         res = self.store.batch([
             {"verb": "delete", "kind": "ConfigMap", "name": alloc["podName"],

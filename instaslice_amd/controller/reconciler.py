@@ -649,6 +649,14 @@ class PodController:
             # drops the pod once no finalizers remain
             reqs = []
             found = self._find_allocation(uid, self._crs())
+            if found is None and FINALIZER_NAME in (md.get("finalizers") or []):
+                # The informer cache can lag the allocation we ourselves
+                # wrote (list+watch stores deliver events asynchronously;
+                # K8sStore tier caught this): stripping the finalizer on a
+                # stale miss orphans the allocation forever — the pod is
+                # gone and nothing re-reconciles it. Confirm against the
+                # source of truth before concluding there is no cleanup.
+                found = self._find_allocation(uid, self.store.list("Instaslice"))
             if found:
                 cr, alloc = found
                 if not (cr["spec"].get("agentManagedTeardown")):
@@ -660,10 +668,6 @@ class PodController:
                     # the two-phase deleted-status handoff (saves one
                     # cross-process event hop + one round-trip per drain)
                     node = cr["metadata"]["name"]
-                    part_uuids = [
-                        k for k, v in (cr["spec"].get("prepared") or {}).items()
-                        if v["podUUID"] == uid
-                    ]
                     reqs.append(
                         {"verb": "delete", "kind": "ConfigMap",
                          "name": alloc["podName"],
@@ -687,10 +691,14 @@ class PodController:
                               "path": ["spec", "gpus", alloc["gpuUUID"],
                                        "usedOrdinals"],
                               "value": alloc["ordinal"]},
-                         ] + [
-                             {"op": "delete",
-                              "path": ["spec", "prepared", puid]}
-                             for puid in part_uuids
+                             # predicate delete on the FRESH object: a
+                             # key list computed from the cached CR can
+                             # miss a prepared entry committed between
+                             # our view and this patch (TOCTOU orphan
+                             # caught by the K8sStore behavioral tier)
+                             {"op": "delete_where",
+                              "path": ["spec", "prepared"],
+                              "field": "podUUID", "value": uid},
                          ]})
                 elif alloc["allocationStatus"] != AllocationStatus.DELETED:
                     reqs.append(

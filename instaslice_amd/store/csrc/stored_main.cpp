@@ -514,6 +514,20 @@ class Store {
           if (!deep_equal(e, v)) list.a->push_back(e);
         node->setkey(leaf, std::move(list));
       }
+    } else if (kind_op == "delete_where") {
+      // predicate delete on the FRESH object (mirror of memstore):
+      // drop map entries whose entry[field] == value
+      const Value* fieldv = op.find("field");
+      Value* cur = node->find(leaf);
+      if (fieldv && fieldv->is_str() && cur && cur->is_map()) {
+        Value kept = Value::map();
+        for (const auto& kv : *cur->m) {
+          const Value* got = kv.second.is_map() ? kv.second.find(fieldv->s)
+                                                : nullptr;
+          if (!(got && deep_equal(*got, v))) kept.setkey(kv.first, kv.second);
+        }
+        node->setkey(leaf, std::move(kept));
+      }
     } else {
       throw StoreError{"Error", "unknown patch op " + kind_op};
     }

@@ -64,6 +64,9 @@ def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
       {"op":"delete", "path":[...]}              remove key if present
       {"op":"add_to_set",      "path":[...], "value":v}  sorted-list add
       {"op":"remove_from_set", "path":[...], "value":v}  sorted-list drop
+      {"op":"delete_where", "path":[...], "field":f, "value":v}
+          drop map entries at path whose entry[f] == v (predicate delete
+          against the fresh object — immune to stale-view key lists)
       {"op":"test",   "path":[...], "value":v}   require equality
       {"op":"test",   "path":[...], "absent":true}  require key missing
 
@@ -120,6 +123,22 @@ def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
             cur = node.get(leaf)
             if isinstance(cur, list) and op["value"] in cur:
                 node[leaf] = [x for x in cur if x != op["value"]]
+        elif kind_op == "delete_where":
+            # predicate delete against the FRESH object: drop map entries
+            # whose entry[field] == value. Exists because key-based deletes
+            # computed from a stale informer view orphan entries committed
+            # between the view and the patch (the teardown prepared-entry
+            # TOCTOU found by the K8sStore behavioral tier).
+            cur = node.get(leaf)
+            if isinstance(cur, dict):
+                field, value = op["field"], op["value"]
+                # field must be PRESENT and equal (a missing field never
+                # matches, even for value None — C++ parity)
+                node[leaf] = {
+                    k: e for k, e in cur.items()
+                    if not (isinstance(e, dict) and field in e
+                            and e[field] == value)
+                }
         else:
             raise ValueError(f"unknown patch op {kind_op!r}")
     return obj

@@ -484,6 +484,8 @@ _ops = st.lists(st.one_of(
                            "value": st.integers(0, 5)}),
     st.fixed_dictionaries({"op": st.just("remove_from_set"), "path": _paths,
                            "value": st.integers(0, 5)}),
+    st.fixed_dictionaries({"op": st.just("delete_where"), "path": _paths,
+                           "field": _keys, "value": _vals}),
     st.fixed_dictionaries({"op": st.just("test"), "path": _paths,
                            "value": _vals}),
     st.fixed_dictionaries({"op": st.just("test"), "path": _paths,

@@ -445,6 +445,11 @@ def main() -> int:
             "vs_baseline": None,  # reference publishes no numbers
             "dtype": "fp32",
             "data": "synthetic",
+            # short windows under-report by ~30% (startup-adjacent jitter;
+            # see --steps help): flag so readers weight the number right.
+            # Driver invocations with --steps 20 land here — quote those
+            # numbers WITH this flag, not alongside full-window ones.
+            "short_window": args.steps < 100,
             "max_rank_rss_mb": round(rss_mb, 1),
             "p50_alloc_latency_ms": round(p50, 3) if p50 is not None else None,
             "p99_alloc_latency_ms": round(p99, 3) if p99 is not None else None,

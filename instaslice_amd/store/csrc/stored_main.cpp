@@ -41,6 +41,7 @@
 #include <csignal>
 #include <cstring>
 #include <deque>
+#include <tuple>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -335,8 +336,13 @@ class Store {
         notify_locked("MODIFIED", repl, wake);
       }
     } else {
+      // DELETED gets its own fresh rv: every event carries a UNIQUE
+      // monotone resourceVersion — the watch resume token (memstore parity)
+      Value obj = *stored;
+      bump_rv(obj);
+      ObjPtr gone = std::make_shared<const Value>(std::move(obj));
       objects_.erase(it);
-      notify_locked("DELETED", stored, wake);
+      notify_locked("DELETED", gone, wake);
     }
     }
     for (Conn* c : wake) c->out_cv.notify_one();
@@ -366,13 +372,37 @@ class Store {
     return out;
   }
 
-  // watch registration: replay + subscribe atomically under the store lock
-  void add_watch(std::shared_ptr<WatchSub> sub, bool replay) {
+  // watch registration: replay/resume + subscribe atomically under the
+  // store lock. `respond` is invoked UNDER the lock, before any event is
+  // enqueued, so the subscribe response (with rev/resumed) precedes every
+  // event on the wire. With has_since and a covered history window, only
+  // the missed events replay (resume tokens); a
+  // compacted window falls back to the full ADDED relist.
+  template <typename RespondFn>
+  void add_watch(std::shared_ptr<WatchSub> sub, bool replay,
+                 bool has_since, uint64_t since, RespondFn respond) {
     Conn* conn = sub->conn;
     bool pushed = false;
     {
       std::lock_guard<std::mutex> g(mu_);
-      if (replay) {
+      bool resumed = false;
+      if (has_since) {
+        bool covered = history_.empty()
+                           ? (since >= rv_)
+                           : (since + 1 >= std::get<0>(history_.front()));
+        resumed = covered;
+      }
+      respond(rv_, resumed);
+      if (resumed) {
+        for (const auto& rec : history_) {
+          if (std::get<0>(rec) > since && matches(*sub, *std::get<2>(rec))) {
+            conn->enqueue_event_silent(sub->watch_id,
+                                       std::get<1>(rec).c_str(),
+                                       std::get<2>(rec));
+            pushed = true;
+          }
+        }
+      } else if (replay) {
         for (const auto& kv : objects_) {
           if (matches(*sub, *kv.second)) {
             conn->enqueue_event_silent(sub->watch_id, "ADDED", kv.second);
@@ -573,6 +603,10 @@ class Store {
   // and the futex wakes are deferred to after the store mutex is released
   void notify_locked(const char* type, const ObjPtr& obj,
                      std::vector<Conn*>& wake) {
+    // bounded event history for watch resume (rv_ was bumped by the
+    // mutation that triggers this notify, so it is the event's revision)
+    history_.emplace_back(rv_, std::string(type), obj);
+    if (history_.size() > kHistoryMax) history_.pop_front();
     bool any_dead = false;
     for (const auto& w : watches_) {
       if (w->dead.load()) { any_dead = true; continue; }
@@ -592,6 +626,8 @@ class Store {
   std::mutex mu_;
   std::map<std::string, ObjPtr> objects_;
   uint64_t rv_ = 0;
+  static constexpr size_t kHistoryMax = 8192;
+  std::deque<std::tuple<uint64_t, std::string, ObjPtr>> history_;
   std::vector<std::shared_ptr<WatchSub>> watches_;
 
   // -- persistence (checkpoint/resume; msgpack snapshot, write-behind) ----
@@ -796,15 +832,25 @@ void reader_loop(Store& store, std::shared_ptr<Conn> conn) {
         }
         const Value* replayv = req.find("replay");
         bool replay = !replayv || replayv->truthy();
-        // respond first so the watch_id precedes any replay event on the
-        // wire: inline send happens-before add_watch enqueues the replay
-        Value result = Value::map();
-        result.setkey("watch_id", Value::integer(sub->watch_id));
-        std::string payload;
-        pack(ok_response(rid, std::move(result)), payload);
-        conn->send_now(frame(payload));
+        const Value* sincev = req.find("since");
+        bool has_since = sincev && sincev->t == Value::T::Int &&
+                         sincev->i >= 0;
+        uint64_t since = has_since ? static_cast<uint64_t>(sincev->i) : 0;
         conn->subs.push_back(sub);
-        store.add_watch(sub, replay);
+        int64_t wid = sub->watch_id;
+        Conn* cptr = conn.get();
+        // respond inside add_watch's lock, before any event is enqueued:
+        // the watch_id + rev/resumed precede every event on the wire
+        store.add_watch(sub, replay, has_since, since,
+                        [&](uint64_t rev, bool resumed) {
+          Value result = Value::map();
+          result.setkey("watch_id", Value::integer(wid));
+          result.setkey("rev", Value::integer(static_cast<int64_t>(rev)));
+          result.setkey("resumed", Value::boolean(resumed));
+          std::string payload;
+          pack(ok_response(rid, std::move(result)), payload);
+          cptr->send_now(frame(payload));
+        });
         continue;
       }
       if (verb == "batch") {

@@ -136,14 +136,18 @@ class _Handler(socketserver.StreamRequestHandler):
                         res = store.batch(req["requests"], quiet=quiet)
                     elif verb == "watch":
                         w = store.watch(req.get("kind"), replay=req.get("replay", True),
-                                        filters=req.get("filters"))
+                                        filters=req.get("filters"),
+                                        since=req.get("since"))
                         watches.append(w)
                         next_watch_id += 1
                         wid = next_watch_id
                         # respond BEFORE pumping so the id precedes any event
                         # on the wire (the client still buffers orphans in
                         # case its caller hasn't registered the id yet)
-                        send({"id": rid, "ok": True, "result": {"watch_id": wid}})
+                        send({"id": rid, "ok": True, "result": {
+                            "watch_id": wid,
+                            "rev": getattr(w, "rev", None),
+                            "resumed": getattr(w, "resumed", False)}})
                         threading.Thread(
                             target=pump_watch, args=(wid, w), daemon=True
                         ).start()
@@ -205,6 +209,10 @@ class _ClientWatch:
         self.kind: Optional[str] = None
         self._cb = None
         self._cb_lock = threading.Lock()
+        # highest event resourceVersion seen — the watch RESUME TOKEN a
+        # reconnect passes as `since` so missed events replay without a
+        # full relist (None until the subscribe response seeds it)
+        self.last_rev: Optional[int] = None
 
     def set_callback(self, fn) -> None:
         with self._cb_lock:
@@ -220,6 +228,12 @@ class _ClientWatch:
     def _deliver(self, event) -> None:
         if self._stopped:
             return
+        try:
+            rv = int(event[1]["metadata"]["resourceVersion"])
+            if self.last_rev is None or rv > self.last_rev:
+                self.last_rev = rv
+        except (KeyError, TypeError, ValueError, IndexError):
+            pass
         with self._cb_lock:
             if self._cb is not None:
                 try:
@@ -360,7 +374,12 @@ class NetStoreClient:
                 with self._idlock:
                     self._next_id += 1
                     rid = self._next_id
+                # resume from the last seen event when possible: the server
+                # replays only the missed window from its bounded history,
+                # falling back to a full ADDED relist on compaction
                 resync = dict(spec, replay=True)
+                if w.last_rev is not None:
+                    resync["since"] = w.last_rev
                 try:
                     with self._wlock:
                         self._sock.sendall(_pack({"id": rid, "verb": "watch",
@@ -451,6 +470,10 @@ class NetStoreClient:
         spec = {"kind": kind, "replay": replay, "filters": filters}
         res = self._call("watch", kind=kind, replay=replay, filters=filters)
         wid = res["watch_id"]
+        # resume-token baseline: the store revision at subscribe time (a
+        # reconnect before any event then resumes from here, not relists)
+        if isinstance(res, dict) and res.get("rev") is not None:
+            w.last_rev = int(res["rev"])
         with self._watch_reg_lock:
             self._watches[wid] = w
             self._watch_specs[wid] = (w, spec)

@@ -167,6 +167,13 @@ def cmd_daemonset(args) -> int:
         smi = NativeAmdSmi()
     agent = NodeAgent(store, smi, node, reset_mode_on_empty=args.reset_on_empty)
     agent.start()
+    plugin = None
+    if getattr(args, "devplugin", False):
+        from instaslice_amd.devplugin import DevicePluginShim
+
+        plugin = DevicePluginShim(store, smi, node,
+                                  resource=args.devplugin_resource).start()
+        log.info("device-plugin shim advertising %s", args.devplugin_resource)
     srv = None
     if args.metrics_port:
         srv = serve_http(get_metrics(), args.metrics_port)
@@ -174,6 +181,8 @@ def cmd_daemonset(args) -> int:
     log.info("daemonset running on node %s (%s)", node,
              "fake" if args.fake else "amdsmi")
     _wait_forever()
+    if plugin:
+        plugin.stop()
     agent.stop()
     if srv:
         srv.shutdown()
@@ -400,6 +409,11 @@ def main(argv=None) -> int:
     p.add_argument("--metrics-port", type=int, default=8084)
     p.add_argument("--reset-on-empty", action="store_true",
                    help="return drained GPUs to SPX/NPS1 (reference parity)")
+    p.add_argument("--devplugin", action="store_true",
+                   help="also run the device-plugin shim (re-advertises "
+                        "partition devices on node capacity after flips)")
+    p.add_argument("--devplugin-resource", default="amd.com/gpu",
+                   help="extended-resource name the shim advertises")
     p.set_defaults(fn=cmd_daemonset)
 
     p = sub.add_parser("submit", help="submit a gated pod requesting a partition")

@@ -395,19 +395,29 @@ def main() -> int:
     # run lifecycles until warm, as a training bench warms its JIT
     t_init = time.perf_counter()
     i_init = 0
-    while time.perf_counter() - t_init < 2.0 and i_init < 2000:
+    # 6 s closes the measured cold-start gap (557 -> 824 across
+    # back-to-back runs was ~6 s of lifecycles; the old 2 s left the
+    # driver's short windows ~25% under full-window numbers)
+    while time.perf_counter() - t_init < 6.0 and i_init < 5000:
         step(-1 - i_init, record=False)
         i_init += 1
 
     for i in range(args.warmup):
         step(i, record=False)
 
+    # a GC pause inside a 20-step (~30 ms) window costs double-digit
+    # percent; collect now, then keep the collector out of the timed region
+    import gc
+
+    gc.collect()
+    gc.disable()
     sync()
     t_start = time.perf_counter()
     for i in range(args.steps):
         step(args.warmup + i, record=True)
     sync()
     elapsed = time.perf_counter() - t_start
+    gc.enable()
 
     import resource
 

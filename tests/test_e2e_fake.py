@@ -812,3 +812,45 @@ def test_flip_batching_overlaps_mode_sets():
         f"mode sets never overlapped (peak={smi.max_concurrent_sets})")
     # sequential floor would be >= 4 * 0.15 = 0.6s of pure set time
     assert wall < 0.55, f"flip batching ineffective: wall={wall:.2f}s"
+
+
+def test_upsize_served_end_to_end():
+    """A cpx request with no CPX slot and no idle GPU rides a free QPX
+    slot through the WHOLE stack: allocation size=2, prepared 2-XCD/72GB
+    partition, env contract intact (the r2 fragmentation mechanism)."""
+    from instaslice_amd.smi.fake import FakeAmdSmi
+
+    c = Cluster(teardown_grace_s=0.0, policy="packed-fit")
+    c.add_node("node-0", FakeAmdSmi(num_gpus=2, node_name="node-0"))
+    c.start()
+    try:
+        for i in range(8):  # GPU0 -> CPX, full
+            c.submit_pod(f"c{i}", "cpx-1x36")
+        for i in range(8):
+            c.wait_pod_scheduled(f"c{i}")
+        for i in range(3):  # GPU1 -> QPX, 3 of 4 slots
+            c.submit_pod(f"q{i}", "qpx-2x72")
+        for i in range(3):
+            c.wait_pod_scheduled(f"q{i}")
+        # no CPX slot free, no idle GPU: the 9th cpx must upsize into
+        # GPU1's remaining QPX slot instead of going unschedulable
+        c.submit_pod("up", "cpx-1x36")
+        c.wait_pod_scheduled("up")
+        allocs = c.allocations("node-0")
+        up = next(a for a in allocs.values() if a["podName"] == "up")
+        assert up["profile"] == "cpx-1x36"          # what was asked
+        assert up["computeMode"] == "QPX" and up["size"] == 2  # what was given
+        prep = next(p for p in c.prepared("node-0").values()
+                    if p["podUUID"] == up["podUUID"])
+        assert prep["xcds"] == 2 and prep["memoryGB"] == 72
+        env = c.pod_env("up")
+        assert env["INSTASLICE_PARTITION_UUID"]
+        assert env["ROCR_VISIBLE_DEVICES"] == str(prep["deviceIndex"])
+        # teardown frees the QPX slot for a real qpx pod
+        c.delete_pod("up")
+        c.wait_pod_gone("up")
+        c.wait_pod_unallocated("up")
+        c.submit_pod("q3", "qpx-2x72")
+        c.wait_pod_scheduled("q3")
+    finally:
+        c.stop()

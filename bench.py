@@ -231,6 +231,19 @@ def main() -> int:
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
 
+    # stdout carries EXACTLY one JSON line (the driver parses it). Gloo's
+    # C++ context prints a "[Gloo] Rank N is connected..." banner straight
+    # to fd 1, past sys.stdout — divert fd 1 to stderr for the whole run
+    # and restore it only around the final JSON print.
+    stdout_fd = os.dup(1)
+    os.dup2(2, 1)
+
+    def print_json_line(s: str) -> None:
+        os.dup2(stdout_fd, 1)
+        sys.stdout.write(s + "\n")
+        sys.stdout.flush()
+        os.dup2(2, 1)
+
     if world > 1:
         # gloo on purpose: an RCCL process group pins a HIP context per rank,
         # and amdgpu refuses partition mode changes while any process holds
@@ -476,7 +489,7 @@ def main() -> int:
                 "policy": args.policy,
             },
         }
-        print(json.dumps(result), flush=True)
+        print_json_line(json.dumps(result))
 
     # teardown
     if prof_stop is not None:

@@ -183,7 +183,11 @@ def scenario_mixed100(args) -> dict:
 
 def scenario_vllm(args) -> dict:
     """vLLM-shaped deployment pod (reference: samples/vllm_dep.yaml requests
-    mig-3g.20gb; MI355X analog: a 2-XCD/72GB partition)."""
+    mig-3g.20gb; MI355X analog: a 2-XCD/72GB partition). This scenario is
+    SIMULATED (env contract only, FakeAmdSmi); the measured version — a
+    real transformer decode loop inside the allocated partition on metal —
+    is tests/test_serving_payload.py::test_serving_workload_in_allocated_partition
+    (gpu tier)."""
     c = _mk_cluster(num_gpus=8)
     try:
         t0 = time.perf_counter()

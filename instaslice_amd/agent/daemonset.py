@@ -596,7 +596,46 @@ class NodeAgent:
                 self._teardown_allocation(cr, pod_uuid, alloc)
                 if self.reset_mode_on_empty:
                     self._maybe_reset_gpu(alloc["gpuUUID"])
+        self._apply_mode_hints(cr, allocations)
         return Result(requeue_after=requeue)
+
+    def _apply_mode_hints(self, cr: dict, allocations: dict) -> None:
+        """Act on controller drain-time hints (spec.gpus[*].desiredMode):
+        pre-flip an IDLE GPU to the demanded mode so the next request's
+        latency excludes the flip wall time. The hint clears when satisfied,
+        raced by a placement (GPU occupied again), refused by the platform,
+        or targeting an unknown/locked GPU — it must never loop."""
+        for gpu_uuid, gd in (cr.get("spec", {}).get("gpus") or {}).items():
+            if not isinstance(gd, dict):
+                continue
+            want = gd.get("desiredMode")
+            if not want:
+                continue
+            g = self._gpus.get(gpu_uuid)
+            occupied = any(
+                isinstance(a, dict) and a.get("gpuUUID") == gpu_uuid
+                for a in allocations.values()
+            )
+            if g is None or gd.get("modeLocked") or occupied \
+                    or g.compute_mode == want:
+                pass  # clear below
+            else:
+                try:
+                    self._ensure_gpu_mode(cr, gpu_uuid, want, g.memory_mode)
+                    self.log.info("pre-flipped idle gpu %s to %s (drain-time "
+                                  "mode hint)", gpu_uuid[:8], want)
+                except SmiBusy:
+                    continue  # transient; retry on the next event
+                except (SmiNotSupported, SmiError) as e:
+                    self.log.warning("mode hint %s on %s refused: %s",
+                                     want, gpu_uuid[:8], e)
+            try:
+                self.store.patch("Instaslice", self.node_name, INSTASLICE_NS, [
+                    {"op": "delete",
+                     "path": ["spec", "gpus", gpu_uuid, "desiredMode"]},
+                ], quiet=True)
+            except (Conflict, NotFound):
+                pass
 
     # -- delete path ----------------------------------------------------------
 

@@ -69,6 +69,9 @@ GPU_SCHEMA = {
         "usedOrdinals": {"type": "array", "items": _INT},
         "index": _INT,
         "modeLocked": _BOOL,
+        # drain-time mode hint (controller writes, agent pre-flips idle GPUs)
+        "desiredMode": {"type": "string",
+                        "enum": ["SPX", "DPX", "TPX", "QPX", "CPX"]},
     },
 }
 

@@ -95,6 +95,9 @@ class PodController:
         # pods currently marked unschedulable: re-reconciled on ANY node-state
         # change so freed capacity is picked up event-driven, not by polling
         self._unschedulable_keys: set = set()
+        # demand memory for drain-time mode planning: profile -> last time a
+        # pod requesting it went unschedulable (drives desiredMode hints)
+        self._starved_profiles: dict = {}
         # node -> (placements_dict, ProfileCatalog): see _profile_for
         self._catalog_cache: dict = {}
         # pod uid -> (victim_name, t): preemption cooldown (_maybe_preempt)
@@ -309,6 +312,7 @@ class PodController:
         key = (md.get("namespace", "default"), md["name"])
         if unschedulable:
             self._unschedulable_keys.add(key)
+            self._starved_profiles[profile_name] = time.time()
         else:
             self._unschedulable_keys.discard(key)
             # common case: the pod was never marked — skip the
@@ -580,6 +584,50 @@ class PodController:
         get_metrics().allocation("preempted")
         return True
 
+    STARVED_TTL_S = 30.0
+
+    def _maybe_hint_mode(self, cr: dict, uid: str, alloc: dict) -> None:
+        """DRAIN-TIME MODE PLANNING: when a teardown
+        leaves a GPU empty while some profile recently went unschedulable,
+        hint the agent to pre-flip the GPU to the starved mode
+        (spec.gpus[uuid].desiredMode). On real hardware a compute-partition
+        set costs O(100ms-1s); pre-flipping during idle time hides that
+        from the next request's allocation latency. The agent only acts on
+        the hint while the GPU is actually idle, so a racing placement
+        simply wins and clears it."""
+        if not self._starved_profiles:
+            return
+        now = time.time()
+        for p, t in list(self._starved_profiles.items()):
+            if now - t > self.STARVED_TTL_S:
+                del self._starved_profiles[p]
+        if not self._starved_profiles:
+            return
+        gpu_uuid = alloc["gpuUUID"]
+        spec = cr["spec"]
+        if any(isinstance(a, dict) and a.get("gpuUUID") == gpu_uuid
+               for k, a in (spec.get("allocations") or {}).items() if k != uid):
+            return  # GPU not drained empty
+        gd = (spec.get("gpus") or {}).get(gpu_uuid) or {}
+        if gd.get("modeLocked"):
+            return
+        prof_name = max(self._starved_profiles,
+                        key=self._starved_profiles.get)
+        profile = self._profile_for(cr, prof_name)
+        if profile is None or profile.compute.value == gd.get("computeMode"):
+            return
+        from instaslice_amd.store.memstore import Conflict
+
+        try:
+            self.store.patch("Instaslice", cr["metadata"]["name"],
+                             INSTASLICE_NS, [
+                {"op": "set",
+                 "path": ["spec", "gpus", gpu_uuid, "desiredMode"],
+                 "value": profile.compute.value},
+            ], quiet=True)
+        except (Conflict, NotFound):
+            pass
+
     def _set_allocation_status(self, cr_name: str, pod_uid: str, status: str,
                                expect: Optional[str] = None) -> None:
         from instaslice_amd.store.memstore import Conflict
@@ -740,6 +788,7 @@ class PodController:
                     emit(self.store, pod, "PartitionReleased",
                          f"partition ordinal {alloc['ordinal']} on "
                          f"{alloc['gpuUUID'][:8]} released")
+                    self._maybe_hint_mode(cr, uid, alloc)
             return Result()
 
         if not pod_is_gated(pod):

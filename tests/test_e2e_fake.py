@@ -854,3 +854,59 @@ def test_upsize_served_end_to_end():
         c.wait_pod_scheduled("q3")
     finally:
         c.stop()
+
+
+def test_drain_time_mode_hint_preflips_idle_gpu():
+    """Drain-time mode planning: a dpx pod goes unschedulable (all GPUs
+    busy in other modes); when a GPU then drains empty, the controller
+    hints desiredMode and the agent PRE-FLIPS it while idle — so the next
+    dpx request's latency excludes the (slow) flip wall time."""
+    from instaslice_amd.smi.fake import FakeAmdSmi
+    from instaslice_amd.controller.reconciler import INSTASLICE_NS
+
+    smi = FakeAmdSmi(num_gpus=1, node_name="node-0",
+                     compute_set_latency_s=0.2)
+    c = Cluster(teardown_grace_s=0.0, policy="packed-fit")
+    c.add_node("node-0", smi)
+    c.start()
+    try:
+        c.submit_pod("small", "cpx-1x36")
+        c.wait_pod_scheduled("small")
+        # dpx demand arrives while the only GPU is CPX-occupied
+        c.submit_pod("big", "dpx-4x144")
+        assert c.wait_pod_outcome("big") == "unschedulable"
+        c.delete_pod("big")
+        c.wait_pod_gone("big")
+        # drain the small pod: GPU empties; hint + pre-flip should follow
+        c.delete_pod("small")
+        c.wait_pod_gone("small")
+        c.wait_allocations_empty("node-0")
+        deadline = time.monotonic() + 5.0
+        while time.monotonic() < deadline:
+            g = smi.list_gpus()[0]
+            if g.compute_mode == "DPX":
+                break
+            time.sleep(0.01)
+        else:
+            cr = c.store.get("Instaslice", "node-0", INSTASLICE_NS)
+            raise AssertionError(
+                f"idle GPU never pre-flipped to DPX; gpus={cr['spec']['gpus']}")
+        # hint must be cleared (no loops)
+        deadline = time.monotonic() + 5.0
+        while time.monotonic() < deadline:
+            cr = c.store.get("Instaslice", "node-0", INSTASLICE_NS)
+            gd = next(iter(cr["spec"]["gpus"].values()))
+            if "desiredMode" not in gd:
+                break
+            time.sleep(0.01)
+        else:
+            raise AssertionError("desiredMode hint never cleared")
+        # the next dpx request pays NO flip: latency well under the 200ms
+        # set latency
+        t0 = time.monotonic()
+        c.submit_pod("big2", "dpx-4x144")
+        c.wait_pod_scheduled("big2")
+        dt = time.monotonic() - t0
+        assert dt < 0.15, f"dpx placement paid the flip anyway ({dt:.3f}s)"
+    finally:
+        c.stop()

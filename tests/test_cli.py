@@ -257,3 +257,27 @@ def test_cli_top_describe_cordon():
             except subprocess.TimeoutExpired:
                 p.kill()
         server.stop()
+
+
+def test_leader_crash_failover_after_ttl():
+    """UNGRACEFUL leader death (no release): the successor must acquire
+    after the lease TTL expires — the crash-failover path, distinct from
+    the graceful-handoff test above."""
+    import time
+
+    from instaslice_amd.store.memstore import MemStore
+
+    store = MemStore()
+    a = LeaderLease(store, "crash-test", identity="A", ttl_s=0.5,
+                    renew_every_s=0.1)
+    # A acquires once and then "crashes": no renew loop ever runs
+    assert a._try_acquire()
+    b = LeaderLease(store, "crash-test", identity="B", ttl_s=0.5,
+                    renew_every_s=0.1).start()
+    try:
+        assert not b.wait_leader(0.3), "B stole a live lease before TTL"
+        assert b.wait_leader(3.0), "B never took over after the TTL expired"
+        lease = store.get("Lease", "crash-test", "instaslice-system")
+        assert lease["spec"]["holderIdentity"] == "B"
+    finally:
+        b.stop()

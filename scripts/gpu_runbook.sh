@@ -18,9 +18,9 @@ timeout 280 python -m instaslice_amd controlplane --port 7080 --grace 0 >> $OUT 
 CP=$!
 sleep 3
 
-log "python -m instaslice_amd daemonset --store 127.0.0.1:7080 --node-name mi355x-node --metrics-port 18084 &"
+log "python -m instaslice_amd daemonset --store 127.0.0.1:7080 --node-name mi355x-node --metrics-port 18084 --devplugin &"
 timeout 270 python -m instaslice_amd daemonset --store 127.0.0.1:7080 \
-  --node-name mi355x-node --metrics-port 18084 >> $OUT 2>&1 &
+  --node-name mi355x-node --metrics-port 18084 --devplugin >> $OUT 2>&1 &
 DS=$!
 sleep 5
 
@@ -59,6 +59,20 @@ EOF
 )
 ROCR_VISIBLE_DEVICES=$DEV timeout 60 ./instaslice_amd/bin/instaslice-payload vecadd 4194304 >> $OUT 2>&1
 ROCR_VISIBLE_DEVICES=$DEV timeout 60 ./instaslice_amd/bin/instaslice-payload info >> $OUT 2>&1
+
+log "device-plugin shim advertisement (amd.com/gpu on node capacity)"
+timeout 30 python - <<'EOF2' >> $OUT 2>&1
+from instaslice_amd.store.netstore import NetStoreClient
+c = NetStoreClient("127.0.0.1", 7080)
+node = c.get("Node", "mi355x-node", "")
+print("node capacity:", node["status"].get("capacity"))
+c.close()
+EOF2
+
+log "SERVING WORKLOAD inside the partition (bf16 transformer decode)"
+ROCR_VISIBLE_DEVICES=$DEV HIP_VISIBLE_DEVICES=$DEV timeout 240 \
+  python -m instaslice_amd.ops.serving_check --layers 4 --dmodel 1024 \
+  --prefill 256 --decode 16 >> $OUT 2>&1
 
 log "curl the agent metrics endpoint"
 timeout 10 curl -s http://127.0.0.1:18084/metrics | grep -E "instaslice_(alloc|partition|reconcile)" | head -12 >> $OUT 2>&1

@@ -183,19 +183,79 @@ PAYLOAD_BIN = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                            "instaslice_amd", "bin", "instaslice-payload")
 
 
-def run_payload(pod_env: dict) -> None:
-    """Run the vecadd validation payload inside the pod's partition, as a
-    child process with the pod's visible-devices env (a real pod's view).
-    The child exits afterwards, so the GPU stays idle for mode flips."""
+class PayloadPool:
+    """Warm payload workers, one per partition device: `instaslice-payload
+    serve` keeps the HIP context alive so a kernel runs inside the pod's
+    partition per request at ~us dispatch cost instead of ~320 ms HIP init
+    per one-shot child (measured, profiles/README.md). A live worker HOLDS
+    its device, which blocks partition mode flips — so the pool is only
+    used under static partitioning; live-partitioning runs keep the
+    one-shot child path (device released between pods)."""
+
+    def __init__(self) -> None:
+        import subprocess as _sp
+
+        self._sp = _sp
+        self._workers: dict = {}
+
+    def _worker(self, device: str):
+        w = self._workers.get(device)
+        if w is not None and w.poll() is None:
+            return w
+        env = dict(os.environ, ROCR_VISIBLE_DEVICES=device)
+        w = self._sp.Popen([PAYLOAD_BIN, "serve"], stdin=self._sp.PIPE,
+                           stdout=self._sp.PIPE, text=True, bufsize=1,
+                           env=env)
+        self._workers[device] = w
+        return w
+
+    def run(self, device: str, cmd: str = "vecadd 1048576") -> None:
+        w = self._worker(device)
+        try:
+            w.stdin.write(cmd + "\n")
+            w.stdin.flush()
+            line = w.stdout.readline()
+        except (BrokenPipeError, OSError) as e:
+            self._workers.pop(device, None)
+            raise RuntimeError(f"payload worker died on device {device}: {e}")
+        res = json.loads(line) if line.startswith("{") else {"ok": False,
+                                                             "error": line}
+        if not res.get("ok"):
+            raise RuntimeError(
+                f"payload FAILED in partition {device}: {res}")
+
+    def close(self) -> None:
+        for w in self._workers.values():
+            try:
+                w.stdin.write("quit\n")
+                w.stdin.flush()
+                w.wait(timeout=5)
+            except Exception:  # noqa: BLE001
+                w.kill()
+        self._workers.clear()
+
+
+_payload_pool: PayloadPool = PayloadPool()
+
+
+def run_payload(pod_env: dict, warm_pool: bool = False) -> None:
+    """Run the vecadd validation payload inside the pod's partition with the
+    pod's visible-devices env (a real pod's view). warm_pool reuses a
+    persistent per-device worker (static partitioning only); otherwise a
+    one-shot child exits afterwards so the GPU stays idle for mode flips."""
+    device = pod_env["ROCR_VISIBLE_DEVICES"]
+    if warm_pool:
+        _payload_pool.run(device)
+        return
     import subprocess
 
     env = dict(os.environ)
-    env["ROCR_VISIBLE_DEVICES"] = pod_env["ROCR_VISIBLE_DEVICES"]
+    env["ROCR_VISIBLE_DEVICES"] = device
     out = subprocess.run([PAYLOAD_BIN, "vecadd", str(1 << 20)],
                          capture_output=True, text=True, timeout=120, env=env)
     if out.returncode != 0:
         raise RuntimeError(
-            f"payload FAILED in partition {pod_env['ROCR_VISIBLE_DEVICES']}: "
+            f"payload FAILED in partition {device}: "
             f"{out.stdout} {out.stderr}")
 
 
@@ -383,7 +443,10 @@ def main() -> int:
         )
         if run_pl:
             cm = store.get("ConfigMap", name, "default")
-            run_payload(cm["data"])
+            # warm pool under static partitioning (no flips to block);
+            # one-shot children otherwise so the device frees between pods
+            run_payload(cm["data"],
+                        warm_pool=partitioning.startswith("static"))
         store.delete("Pod", name, "default")
 
         state = {"pod_gone": False, "alloc_gone": False}
@@ -492,6 +555,7 @@ def main() -> int:
         print_json_line(json.dumps(result))
 
     # teardown
+    _payload_pool.close()
     if prof_stop is not None:
         prof_stop()
     events.stop()

@@ -10,12 +10,22 @@
 //   instaslice-payload membw [BYTES] [ITERS]
 //   instaslice-payload busy [MS]
 //   instaslice-payload census
+//   instaslice-payload serve      (persistent worker: commands on stdin,
+//                                  one JSON line per command on stdout —
+//                                  HIP init is paid once, so a warm pool
+//                                  can run a kernel per pod lifecycle at
+//                                  production rate; NOTE the live process
+//                                  holds the device, which blocks
+//                                  partition mode flips: serve is for
+//                                  static-partitioning phases only)
 //
 // Build: see build_native.py (hipcc --offload-arch=gfx950).
 
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <iostream>
+#include <sstream>
 #include <string>
 
 #include "payload_core.hpp"
@@ -65,6 +75,39 @@ int main(int argc, char** argv) {
       for (unsigned v : c)
         if (v) ++xcds;
       std::printf("], \"xcds_visible\": %d}\n", xcds);
+    } else if (std::strcmp(cmd, "serve") == 0) {
+      // warm-pool worker: one line per request, one JSON line per reply
+      //   vecadd <n> | busy <ms> | ping | quit
+      std::string line;
+      while (std::getline(std::cin, line)) {
+        std::istringstream iss(line);
+        std::string verb;
+        iss >> verb;
+        if (verb.empty()) continue;
+        try {
+          if (verb == "quit") { std::printf("{\"ok\": true}\n"); break; }
+          if (verb == "ping") {
+            std::printf("{\"ok\": true, \"cmd\": \"ping\"}\n");
+          } else if (verb == "vecadd") {
+            size_t n = 1 << 20;
+            iss >> n;
+            double err = run_vecadd(n);
+            std::printf(
+                "{\"ok\": %s, \"cmd\": \"vecadd\", \"n\": %zu, \"max_err\": %g}\n",
+                err == 0.0 ? "true" : "false", n, err);
+          } else if (verb == "busy") {
+            double ms = 10.0;
+            iss >> ms;
+            run_busy(ms);
+            std::printf("{\"ok\": true, \"cmd\": \"busy\", \"ms\": %.1f}\n", ms);
+          } else {
+            std::printf("{\"ok\": false, \"error\": \"unknown verb\"}\n");
+          }
+        } catch (const std::exception& e) {
+          std::printf("{\"ok\": false, \"error\": \"%s\"}\n", e.what());
+        }
+        std::fflush(stdout);
+      }
     } else {
       std::fprintf(stderr, "unknown command: %s\n", cmd);
       return 2;

@@ -73,3 +73,26 @@ def test_serving_workload_in_allocated_partition():
         c.wait_pod_gone("vllm-0", namespace="serving", timeout=30.0)
     finally:
         c.stop()
+
+
+def test_payload_serve_protocol_cpu():
+    """The serve-mode line protocol itself (ping/quit/unknown) works
+    without touching HIP — CPU-tier coverage of the worker loop."""
+    bin_path = ROOT / "instaslice_amd" / "bin" / "instaslice-payload"
+    if not bin_path.exists():
+        pytest.skip("instaslice-payload not built")
+    w = subprocess.Popen([str(bin_path), "serve"], stdin=subprocess.PIPE,
+                         stdout=subprocess.PIPE, text=True, bufsize=1)
+    try:
+        def ask(cmd):
+            w.stdin.write(cmd + "\n")
+            w.stdin.flush()
+            return json.loads(w.stdout.readline())
+
+        assert ask("ping")["ok"]
+        assert not ask("nonsense")["ok"]
+        assert ask("quit")["ok"]
+        assert w.wait(timeout=10) == 0
+    finally:
+        if w.poll() is None:
+            w.kill()

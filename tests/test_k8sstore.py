@@ -380,3 +380,46 @@ def test_k8s_heartbeat_lands_in_status(k8s_cluster):
             return
         time.sleep(0.1)
     pytest.fail("agent heartbeat never landed in CR status over K8sStore")
+
+
+def test_k8s_randomized_lifecycle_invariants():
+    """The randomized invariant sweep (tests/test_invariants_e2e.py) over
+    the Kubernetes adapter: RFC3339 timestamps, status subresources and
+    client-side patch emulation must preserve the same global consistency
+    under interleaved submit/delete churn."""
+    import random
+
+    from instaslice_amd.runtime.cluster import Cluster
+    from instaslice_amd.smi import FakeAmdSmi
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent))
+    from test_invariants_e2e import _check_invariants, _quiesce
+
+    kubernetes.reset_server()
+    rng = random.Random(5)
+    c = Cluster(store=K8sStore(), teardown_grace_s=0.0)
+    c.add_node("node-0", FakeAmdSmi(num_gpus=2, node_name="node-0"))
+    c.start()
+    live, counter = set(), 0
+    try:
+        for _ in range(4):
+            for _ in range(rng.randint(3, 8)):
+                if live and rng.random() < 0.45:
+                    victim = rng.choice(sorted(live))
+                    live.discard(victim)
+                    c.delete_pod(victim)
+                else:
+                    name = f"k8s-p{counter}"
+                    counter += 1
+                    c.submit_pod(name, rng.choice(
+                        ["cpx-1x36", "qpx-2x72", "dpx-4x144"]))
+                    live.add(name)
+            _quiesce(c, live, timeout=40.0)
+            _check_invariants(c, "node-0")
+        for name in sorted(live):
+            c.delete_pod(name)
+        live.clear()
+        _quiesce(c, live, timeout=40.0)
+        _check_invariants(c, "node-0")
+    finally:
+        c.stop()

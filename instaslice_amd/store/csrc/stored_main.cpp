@@ -73,15 +73,39 @@ struct WatchSub {
   std::atomic<bool> dead{false};
 };
 
+// One watch event shared by every subscribed connection. The msgpack body
+// `[type, obj]` is encoded EXACTLY ONCE (first writer thread to flush it),
+// then every connection splices its own {watch_id, event} header around
+// the shared bytes — at w8 each CR event fans out to 8+ kind-wide watches
+// and per-connection re-encoding of the same ~KB object was the store
+// daemon's dominant CPU (measured ~3.6 cores; profiles/stored_cpu_w8.log).
+struct EventBlob {
+  std::string ev_type;
+  std::shared_ptr<const Value> obj;
+  std::once_flag once;
+  std::string body;  // msgpack [type, obj]
+
+  EventBlob(const char* t, std::shared_ptr<const Value> o)
+      : ev_type(t), obj(std::move(o)) {}
+
+  const std::string& ensure_encoded() {
+    std::call_once(once, [this] {
+      body.push_back(static_cast<char>(0x92));
+      pack_str(ev_type, body);
+      pack(*obj, body);
+    });
+    return body;
+  }
+};
+
 // per-connection outbox: responses and watch events share one ordered queue
 // flushed by a dedicated writer thread, so a slow consumer never blocks the
 // store mutex or another connection. Watch events enqueue as (watch_id,
-// shared immutable Value) and are packed by the writer thread.
+// shared EventBlob) and are packed by the writer thread.
 struct OutItem {
   std::string bytes;               // pre-framed (responses)
   int64_t wid = -1;                // >= 0: pack {watch_id, event} lazily
-  const char* ev_type = nullptr;
-  std::shared_ptr<const Value> ev_obj;
+  std::shared_ptr<EventBlob> blob;
 };
 
 struct Conn {
@@ -137,15 +161,13 @@ struct Conn {
     out_cv.notify_one();
   }
 
-  void enqueue_event(int64_t wid, const char* type,
-                     std::shared_ptr<const Value> obj) {
+  void enqueue_event(int64_t wid, std::shared_ptr<EventBlob> blob) {
     {
       std::lock_guard<std::mutex> g(out_mu);
       if (closing) return;
       OutItem it;
       it.wid = wid;
-      it.ev_type = type;
-      it.ev_obj = std::move(obj);
+      it.blob = std::move(blob);
       outbox.push_back(std::move(it));
     }
     out_cv.notify_one();
@@ -160,8 +182,7 @@ struct Conn {
   // reconnect-enabled client resyncs with replay).
   static constexpr size_t kMaxOutbox = 65536;
 
-  void enqueue_event_silent(int64_t wid, const char* type,
-                            std::shared_ptr<const Value> obj) {
+  void enqueue_event_silent(int64_t wid, std::shared_ptr<EventBlob> blob) {
     std::lock_guard<std::mutex> g(out_mu);
     if (closing) return;
     if (outbox.size() >= kMaxOutbox) {
@@ -172,8 +193,7 @@ struct Conn {
     }
     OutItem it;
     it.wid = wid;
-    it.ev_type = type;
-    it.ev_obj = std::move(obj);
+    it.blob = std::move(blob);
     outbox.push_back(std::move(it));
   }
 };
@@ -396,16 +416,19 @@ class Store {
       if (resumed) {
         for (const auto& rec : history_) {
           if (std::get<0>(rec) > since && matches(*sub, *std::get<2>(rec))) {
-            conn->enqueue_event_silent(sub->watch_id,
-                                       std::get<1>(rec).c_str(),
-                                       std::get<2>(rec));
+            conn->enqueue_event_silent(
+                sub->watch_id,
+                std::make_shared<EventBlob>(std::get<1>(rec).c_str(),
+                                            std::get<2>(rec)));
             pushed = true;
           }
         }
       } else if (replay) {
         for (const auto& kv : objects_) {
           if (matches(*sub, *kv.second)) {
-            conn->enqueue_event_silent(sub->watch_id, "ADDED", kv.second);
+            conn->enqueue_event_silent(
+                sub->watch_id,
+                std::make_shared<EventBlob>("ADDED", kv.second));
             pushed = true;
           }
         }
@@ -608,10 +631,13 @@ class Store {
     history_.emplace_back(rv_, std::string(type), obj);
     if (history_.size() > kHistoryMax) history_.pop_front();
     bool any_dead = false;
+    // ONE blob per event: every matching connection shares the encoded body
+    std::shared_ptr<EventBlob> blob;
     for (const auto& w : watches_) {
       if (w->dead.load()) { any_dead = true; continue; }
       if (!matches(*w, *obj)) continue;
-      w->conn->enqueue_event_silent(w->watch_id, type, obj);
+      if (!blob) blob = std::make_shared<EventBlob>(type, obj);
+      w->conn->enqueue_event_silent(w->watch_id, blob);
       if (wake.empty() || wake.back() != w->conn) wake.push_back(w->conn);
     }
     if (any_dead) {
@@ -746,16 +772,18 @@ Value execute(Store& store, Value& req) {
   throw StoreError{"Error", "unknown verb '" + verb + "'"};
 }
 
-// pack one lazily-encoded watch event: {"watch_id": N, "event": [type, obj]}
+// pack one watch event: {"watch_id": N, "event": [type, obj]} — the event
+// body bytes are encoded once per EVENT (EventBlob), shared across every
+// subscribed connection; only the tiny per-watch header is built here.
 std::string pack_event(const OutItem& it) {
+  const std::string& body = it.blob->ensure_encoded();
   std::string payload;
+  payload.reserve(body.size() + 24);
   payload.push_back(static_cast<char>(0x82));
   pack_str("watch_id", payload);
   pack_int(it.wid, payload);
   pack_str("event", payload);
-  payload.push_back(static_cast<char>(0x92));
-  pack_str(it.ev_type, payload);
-  pack(*it.ev_obj, payload);
+  payload += body;
   return frame(payload);
 }
 

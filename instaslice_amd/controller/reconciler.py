@@ -98,6 +98,9 @@ class PodController:
         # demand memory for drain-time mode planning: profile -> last time a
         # pod requesting it went unschedulable (drives desiredMode hints)
         self._starved_profiles: dict = {}
+        # (ns, name) -> last event-driven retry (herd limit; see
+        # _instaslice_to_pods). Entries are pruned with _unschedulable_keys.
+        self._unsched_last_retry: dict = {}
         # node -> (placements_dict, ProfileCatalog): see _profile_for
         self._catalog_cache: dict = {}
         # pod uid -> (victim_name, t): preemption cooldown (_maybe_preempt)
@@ -136,6 +139,14 @@ class PodController:
     # Reference: podMapFunc maps an Instaslice with "created" allocations back
     # to its pods (instaslice_controller.go:398-407).
 
+    # waiting-pod retry herd limit: re-reconciling EVERY unschedulable pod
+    # on EVERY node event is O(waiting x event-rate) — with thousands
+    # waiting that's a reconcile storm (docs/FUTURE.md item). Each waiting
+    # pod gets at most one event-driven retry per window; correctness is
+    # unaffected because placement is re-validated per retry and the
+    # periodic requeue fallback still exists.
+    UNSCHED_RETRY_WINDOW_S = 0.05
+
     def _instaslice_to_pods(self, event_type: str, obj: dict) -> List[Key]:
         keys: List[Key] = []
         for alloc in (obj.get("spec", {}).get("allocations") or {}).values():
@@ -145,7 +156,12 @@ class PodController:
             ) and self._owns(alloc["namespace"], alloc["podName"]):
                 keys.append(("Pod", alloc["namespace"], alloc["podName"]))
         # freed/changed capacity: give waiting pods another placement pass
-        keys.extend(("Pod", ns, name) for (ns, name) in list(self._unschedulable_keys))
+        now = time.monotonic()
+        for (ns, name) in list(self._unschedulable_keys):
+            last = self._unsched_last_retry.get((ns, name), 0.0)
+            if now - last >= self.UNSCHED_RETRY_WINDOW_S:
+                self._unsched_last_retry[(ns, name)] = now
+                keys.append(("Pod", ns, name))
         return keys
 
     # -- helpers ------------------------------------------------------------
@@ -315,6 +331,7 @@ class PodController:
             self._starved_profiles[profile_name] = time.time()
         else:
             self._unschedulable_keys.discard(key)
+            self._unsched_last_retry.pop(key, None)
             # common case: the pod was never marked — skip the
             # read-modify-write entirely (one store GET per placement
             # otherwise; stale-view miss just leaves a cosmetic annotation
@@ -671,6 +688,7 @@ class PodController:
                 pod = self.store.get("Pod", name, namespace)
             except NotFound:
                 self._unschedulable_keys.discard((namespace, name))
+                self._unsched_last_retry.pop((namespace, name), None)
                 return Result()
         md = pod["metadata"]
         uid = md.get("uid")
@@ -683,6 +701,7 @@ class PodController:
         # teardown path (reference: instaslice_controller.go:99-142)
         if md.get("deletionTimestamp"):
             self._unschedulable_keys.discard((namespace, name))
+            self._unsched_last_retry.pop((namespace, name), None)
             self._pending_since.pop(uid, None)  # deleted before ungating
             self._preempted_for.pop(uid, None)  # bounded cooldown map
             gated = pod_is_gated(pod)

@@ -622,6 +622,16 @@ class NodeAgent:
             else:
                 try:
                     self._ensure_gpu_mode(cr, gpu_uuid, want, g.memory_mode)
+                    from instaslice_amd.metrics import get_metrics
+
+                    get_metrics().allocation("preflip")
+                    emit(self.store,
+                         {"kind": "Instaslice",
+                          "namespace": INSTASLICE_NS,
+                          "name": self.node_name},
+                         "PreFlipped",
+                         f"idle GPU {gpu_uuid[:8]} pre-flipped to {want} "
+                         "for starved demand (drain-time mode planning)")
                     self.log.info("pre-flipped idle gpu %s to %s (drain-time "
                                   "mode hint)", gpu_uuid[:8], want)
                 except SmiBusy:

@@ -304,11 +304,18 @@ class PodController:
                     md["name"], profile_name, node_name,
                     placement.gpu_uuid[:8], placement.ordinal,
                 )
+                upsized = actual_xcds > profile.xcds
+                if upsized:
+                    from instaslice_amd.metrics import get_metrics
+
+                    get_metrics().allocation("upsized")
                 emit(self.store, pod, "Placed",
                      f"allocated {profile_name} on {node_name}/"
                      f"{placement.gpu_uuid[:8]}#{placement.ordinal}"
                      + (" (mode change planned)"
-                        if placement.needs_mode_change else ""))
+                        if placement.needs_mode_change else "")
+                     + (f" (upsized to a {actual_xcds}-XCD "
+                        f"{placement.compute_mode} slot)" if upsized else ""))
                 self._mark_unschedulable(md, profile_name, unschedulable=False)
                 # normally the agent's "created" event advances this pod; the
                 # timed recheck only notices a dead agent (no events then)

@@ -10,10 +10,12 @@
 // CPU test tier runs the same battery against both (tests/test_native_store
 // .py) — but under load the daemon is the control plane's scale-out floor:
 // the Python server serializes ~1 ms of GIL-bound wire handling per pod
-// lifecycle, which caps an 8-agent cluster near ~400 pods/s.
+// lifecycle, which caps an 8-agent cluster near ~400 pods/s; this daemon
+// carries the same topology at 3.3-4.2k pods/s (w8, 256-core box, r2).
 //
-// Throughput discipline (the global object-map mutex bounds cluster-wide
-// rate, measured as the w4->w8 plateau at ~1760 pods/s):
+// Throughput discipline (the serialized per-lifecycle event/request chain
+// bounds cluster-wide rate; stored itself measures ~3.8 cores at w8 —
+// profiles/stored_cpu_w8*.log):
 //   - objects are IMMUTABLE once stored (shared_ptr<const Value>); writers
 //     build replacements with copy-on-write along the touched path only —
 //     a patch clones ~hundreds of bytes of a ~3 KB CR, the rest is shared

@@ -908,5 +908,14 @@ def test_drain_time_mode_hint_preflips_idle_gpu():
         c.wait_pod_scheduled("big2")
         dt = time.monotonic() - t0
         assert dt < 0.15, f"dpx placement paid the flip anyway ({dt:.3f}s)"
+        # the pre-flip surfaced as an operator-visible event
+        deadline = time.monotonic() + 5.0
+        while time.monotonic() < deadline:
+            evs = [e for e in c.store.list("Event")
+                   if e.get("reason") == "PreFlipped"]
+            if evs:
+                break
+            time.sleep(0.02)
+        assert evs, "PreFlipped event never recorded"
     finally:
         c.stop()

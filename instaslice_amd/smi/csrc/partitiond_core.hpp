@@ -167,6 +167,17 @@ class Partitiond {
     stale_ = true;
   }
 
+  // The MODE-based memory set (amdsmi.h:5920) — distinct entry point from
+  // the type-based one above; probed separately so the platform status
+  // matrix covers every write variant.
+  void set_memory_partition_mode(uint32_t index, const std::string& mode) {
+    std::lock_guard<std::mutex> lk(mu_);
+    check(amdsmi_set_gpu_memory_partition_mode(handle_locked(index),
+                                               memory_mode_from_name(mode)),
+          "amdsmi_set_gpu_memory_partition_mode");
+    stale_ = true;
+  }
+
   // Accelerator-partition profile catalog (amdsmi.h:5950); the MI355X analog
   // of GetGpuInstanceProfileInfo+PossiblePlacements discovery
   // (instaslice_daemonset.go:613-658).

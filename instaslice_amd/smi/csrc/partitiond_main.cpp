@@ -86,10 +86,12 @@ int main() {
         std::string v = (verb == "get_compute") ? d.get_compute_partition(idx)
                                                 : d.get_memory_partition(idx);
         reply_ok("\"" + json_escape(v) + "\"");
-      } else if (verb == "set_compute" || verb == "set_memory") {
+      } else if (verb == "set_compute" || verb == "set_memory" ||
+                 verb == "set_memory_mode") {
         uint32_t idx; std::string mode; iss >> idx >> mode;
         if (verb == "set_compute") d.set_compute_partition(idx, mode);
-        else d.set_memory_partition(idx, mode);
+        else if (verb == "set_memory") d.set_memory_partition(idx, mode);
+        else d.set_memory_partition_mode(idx, mode);
         reply_ok();
       } else if (verb == "set_profile") {
         uint32_t idx, prof; iss >> idx >> prof;

@@ -55,6 +55,24 @@ def _key(obj: dict) -> Key:
     return (obj["kind"], md.get("namespace", ""), md["name"])
 
 
+def json_equal(a, b) -> bool:
+    """Typed JSON equality: bool is its OWN type (True != 1), numerics
+    cross-compare (1 == 1.0), containers recurse. This is the C++ store's
+    deep_equal semantics — Python's native `==` treats bools as ints,
+    which the differential fuzzer caught as a parity divergence in
+    test-op evaluation."""
+    a_bool, b_bool = isinstance(a, bool), isinstance(b, bool)
+    if a_bool != b_bool:
+        return False
+    if isinstance(a, dict) and isinstance(b, dict):
+        if a.keys() != b.keys():
+            return False
+        return all(json_equal(a[k], b[k]) for k in a)
+    if isinstance(a, list) and isinstance(b, list):
+        return len(a) == len(b) and all(map(json_equal, a, b))
+    return a == b
+
+
 def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
     """Apply patch ops to `obj` in place (shared by MemStore.patch and the
     K8sStore emulation). Ops, applied in order:
@@ -85,7 +103,7 @@ def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
             if op.get("absent"):
                 if not missing:
                     raise Conflict(f"patch test: {path} expected absent")
-            elif missing or node != op.get("value"):
+            elif missing or not json_equal(node, op.get("value")):
                 raise Conflict(f"patch test: {path} != {op.get('value')!r}")
             continue
         # navigate to parent, creating dicts along the way; lists are leaf
@@ -116,13 +134,14 @@ def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
             cur = node.get(leaf)
             if not isinstance(cur, list):
                 cur = []
-            if op["value"] not in cur:
+            if not any(json_equal(x, op["value"]) for x in cur):
                 cur = sorted(cur + [op["value"]])
             node[leaf] = cur
         elif kind_op == "remove_from_set":
             cur = node.get(leaf)
-            if isinstance(cur, list) and op["value"] in cur:
-                node[leaf] = [x for x in cur if x != op["value"]]
+            if isinstance(cur, list):
+                node[leaf] = [x for x in cur
+                              if not json_equal(x, op["value"])]
         elif kind_op == "delete_where":
             # predicate delete against the FRESH object: drop map entries
             # whose entry[field] == value. Exists because key-based deletes
@@ -137,7 +156,7 @@ def apply_patch_ops(obj: dict, ops: List[dict]) -> dict:
                 node[leaf] = {
                     k: e for k, e in cur.items()
                     if not (isinstance(e, dict) and field in e
-                            and e[field] == value)
+                            and json_equal(e[field], value))
                 }
         else:
             raise ValueError(f"unknown patch op {kind_op!r}")

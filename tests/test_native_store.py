@@ -511,7 +511,7 @@ def _fuzz_backends():
 
 @needs_stored
 @settings(max_examples=int(os.environ.get(
-              "INSTASLICE_FUZZ_EXAMPLES", "25")),
+              "INSTASLICE_FUZZ_EXAMPLES", "150")),
           deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(batches=st.lists(_ops, min_size=1, max_size=5))
@@ -560,7 +560,7 @@ _verb_ops = st.lists(st.one_of(
 
 @needs_stored
 @settings(max_examples=int(os.environ.get(
-              "INSTASLICE_FUZZ_EXAMPLES", "25")),
+              "INSTASLICE_FUZZ_EXAMPLES", "150")),
           deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
 @given(seq=_verb_ops)

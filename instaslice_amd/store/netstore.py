@@ -117,7 +117,9 @@ class _Handler(socketserver.StreamRequestHandler):
                     continue
                 try:
                     send({"watch_id": watch_id, "event": list(ev)})
-                except (BrokenPipeError, OSError):
+                except (BrokenPipeError, OSError, ValueError):
+                    # ValueError: wfile closed by the handler thread while a
+                    # severed client tears down — same fate as a broken pipe
                     w.stop()
                     return
 

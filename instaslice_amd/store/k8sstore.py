@@ -187,14 +187,38 @@ class K8sStore:
                 "resourceVersion"]
             return body
 
+        def status_phase(replace_status, res: dict) -> dict:
+            """Chained status replace with ITS OWN conflict retry. The main
+            replace has already committed, so a 409 here (another writer
+            slipped between our two calls) must NOT surface as Conflict —
+            the caller's update_with_retry would re-apply a mutation that
+            already landed (double-applied counters caught by the
+            contention test). Refresh the rv and retry the status write
+            alone (last-writer-wins per subresource, the controller-runtime
+            status().update pattern)."""
+            body = with_rv(res)
+            for _ in range(16):
+                try:
+                    return replace_status(body)
+                except ApiException as e:
+                    if getattr(e, "status", None) != 409:
+                        raise
+                    fresh = self.get(kind, name, ns)
+                    body["metadata"]["resourceVersion"] = fresh["metadata"][
+                        "resourceVersion"]
+            raise Conflict(f"{kind}/{name}: status replace kept conflicting")
+
         try:
             if kind == "Instaslice":
                 res = self._custom.replace_namespaced_custom_object(
                     API_GROUP, API_VERSION, ns, "instaslices", name, obj)
                 if obj.get("status") is not None:
-                    res = self._custom.replace_namespaced_custom_object_status(
-                        API_GROUP, API_VERSION, ns, "instaslices", name,
-                        with_rv(res))
+                    res = status_phase(
+                        lambda b: self._custom
+                        .replace_namespaced_custom_object_status(
+                            API_GROUP, API_VERSION, ns, "instaslices",
+                            name, b),
+                        res)
                 return res
             if kind == "Pod":
                 return self._to_dict(self._core.replace_namespaced_pod(name, ns, obj))
@@ -207,8 +231,10 @@ class K8sStore:
                 # removed capacity keys actually go away
                 res = self._to_dict(self._core.replace_node(name, obj))
                 if obj.get("status") is not None:
-                    res = self._to_dict(
-                        self._core.replace_node_status(name, with_rv(res)))
+                    res = status_phase(
+                        lambda b: self._to_dict(
+                            self._core.replace_node_status(name, b)),
+                        res)
                 return res
             if kind == "Lease":
                 return self._to_dict(
